@@ -1,0 +1,183 @@
+"""Mode-string sockets over the C++ shm ring engine.
+
+Mirrors the reference's swappable-transport seam (uber/fiber
+``fiber/socket.py:323-425``: modes ``"r"/"w"/"rw"/"req"/"rep"``) with
+single-node shared-memory semantics:
+
+* an **address** is a short string (the shm segment base name) — it is
+  picklable and re-dialable from any process on the node (the reference's
+  ``tcp://host:port`` trick, ``fiber/queues.py:122-137``);
+* ``"w"``/``"r"`` — many-producer / many-consumer ends of one MPMC ring
+  (push/pull; demand-driven consumption gives at-least-nanomsg fairness);
+* ``"rw"`` — bidirectional pair of rings (Pipe);
+* ``"req"``/``"rep"`` — credit/request channel: REQ pushes
+  ``ident``-enveloped requests on a shared ring and receives on a private
+  per-ident reply ring; REP pops requests and replies by ident (what the
+  resilient pool's pending-table attribution rides on).
+"""
+
+import atexit
+import os
+import struct
+import threading
+
+from . import config as fam_config
+from . import util
+
+try:
+    from . import _transport
+except ImportError as exc:  # pragma: no cover
+    raise ImportError(
+        "fiber_amd._transport extension is not built; run "
+        "`python setup.py build_ext --inplace`"
+    ) from exc
+
+ShmRing = _transport.ShmRing
+
+_owned_rings = []
+_owned_lock = threading.Lock()
+
+
+def _register_owned(ring):
+    with _owned_lock:
+        _owned_rings.append(ring)
+
+
+@atexit.register
+def _cleanup_owned():
+    with _owned_lock:
+        for ring in _owned_rings:
+            try:
+                ring.close()
+                ring.unlink()
+            except Exception:
+                pass
+        _owned_rings.clear()
+
+
+def new_address(prefix="fam-ch"):
+    return util.random_name(prefix)
+
+
+def _capacity():
+    return fam_config.get_object().ring_capacity
+
+
+def _open(name, create, capacity=None):
+    ring = ShmRing(name, create, capacity or _capacity(), 20.0)
+    if create:
+        _register_owned(ring)
+    return ring
+
+
+class Socket:
+    """A mode-string socket bound or connected to an address."""
+
+    def __init__(self, mode, addr=None, bind=False, capacity=None):
+        if mode not in ("r", "w", "rw", "req", "rep"):
+            raise ValueError("bad socket mode %r" % mode)
+        self.mode = mode
+        self.addr = addr or new_address()
+        self.bound = bind
+        cap = capacity
+        self._rings = {}
+
+        if mode in ("r", "w"):
+            # One shared MPMC ring.  The binder creates it.
+            self._rings["main"] = _open(self.addr, bind, cap)
+        elif mode == "rw":
+            # .a flows binder->dialer, .b flows dialer->binder.
+            self._rings["a"] = _open(self.addr + ".a", bind, cap)
+            self._rings["b"] = _open(self.addr + ".b", bind, cap)
+        elif mode == "rep":
+            if not bind:
+                raise ValueError("'rep' sockets must bind")
+            self._rings["q"] = _open(self.addr + ".q", True, cap)
+            self._reply_cache = {}
+        elif mode == "req":
+            self.ident = util.random_name("i")[:24]
+            self._rings["q"] = _open(self.addr + ".q", False, cap)
+            self._rings["r"] = _open(
+                self.addr + ".r." + self.ident, True, cap
+            )
+
+    # -- plain r/w ---------------------------------------------------------
+    def send(self, data, timeout=-1.0):
+        if self.mode == "w":
+            return self._rings["main"].send(data, timeout)
+        if self.mode == "rw":
+            ring = self._rings["b"] if not self.bound else self._rings["a"]
+            return ring.send(data, timeout)
+        if self.mode == "req":
+            ident = self.ident.encode()
+            envelope = struct.pack(">B", len(ident)) + ident + bytes(data)
+            return self._rings["q"].send(envelope, timeout)
+        raise ValueError("socket mode %r cannot send()" % self.mode)
+
+    def recv(self, timeout=-1.0):
+        if self.mode == "r":
+            return self._rings["main"].recv(timeout)
+        if self.mode == "rw":
+            ring = self._rings["a"] if not self.bound else self._rings["b"]
+            return ring.recv(timeout)
+        if self.mode == "req":
+            return self._rings["r"].recv(timeout)
+        raise ValueError("socket mode %r cannot recv()" % self.mode)
+
+    # -- rep ---------------------------------------------------------------
+    def recv_request(self, timeout=-1.0):
+        """REP side: returns (ident, payload) or None on timeout."""
+        data = self._rings["q"].recv(timeout)
+        if data is None:
+            return None
+        (ilen,) = struct.unpack(">B", data[:1])
+        ident = data[1 : 1 + ilen].decode()
+        return ident, data[1 + ilen :]
+
+    def send_reply(self, ident, data, timeout=-1.0):
+        ring = self._reply_cache.get(ident)
+        if ring is None:
+            ring = _open(self.addr + ".r." + ident, False)
+            self._reply_cache[ident] = ring
+        return ring.send(data, timeout)
+
+    def drop_peer(self, ident):
+        """REP side: forget a dead requester's reply ring."""
+        ring = self._reply_cache.pop(ident, None)
+        if ring is not None:
+            try:
+                ring.detach()
+            except Exception:
+                pass
+
+    # -- stats / lifecycle -------------------------------------------------
+    @property
+    def pending(self):
+        if self.mode in ("r", "w"):
+            return self._rings["main"].size
+        return sum(r.size for r in self._rings.values())
+
+    def close(self):
+        for ring in self._rings.values():
+            try:
+                if ring.is_owner:
+                    ring.close()
+                    ring.unlink()
+                ring.detach()
+            except Exception:
+                pass
+        if self.mode == "rep":
+            for ring in self._reply_cache.values():
+                try:
+                    ring.detach()
+                except Exception:
+                    pass
+            self._reply_cache.clear()
+        self._rings = {}
+
+    def __repr__(self):
+        return "Socket(mode=%r, addr=%r, bound=%r)" % (
+            self.mode,
+            self.addr,
+            self.bound,
+        )
