@@ -1,0 +1,676 @@
+"""Worker pools over the shm transport.
+
+Rebuild target #1: the reference's ``ZPool`` /
+``ResilientZPool`` data path (uber/fiber ``fiber/pool.py:644-1692``),
+re-based on shared-memory rings:
+
+* :class:`ZPool` — master pushes chunked tasks onto one MPMC task ring;
+  workers (fiber Processes, optionally GPU-pinned) pop on demand
+  (demand-driven consumption replaces nanomsg PUSH round-robin — strictly
+  better load balance for uneven task durations) and push chunked results
+  onto a result ring.  Full multiprocessing.Pool API.
+* :class:`ResilientZPool` — REQ/REP credit scheduling: a worker *requests*
+  each chunk, the master records it in a per-worker pending table, and a
+  dead worker's pending chunks are resubmitted after respawn.  Function
+  exceptions kill the worker (tasks must be idempotent) and the chunk is
+  retried, so probabilistic failures still converge to a complete result.
+
+Task wire format: ``(seq, base, func_blob, args, starmap, kwds)``.
+Result wire format: ``(seq, base, values, failure, ident)`` where
+``failure`` is None or ``(index, exception)``.
+"""
+
+import itertools
+import os
+import queue as _stdlib_queue
+import threading
+import time
+
+from . import config as fam_config
+from . import serialization, util
+from .process import Process
+from .transport import Socket
+
+DEFAULT_CHUNKSIZE = 32
+_SENTINEL_SEQ = -1
+
+
+class _ExcInfo:
+    """Picklable carrier for a worker-side exception."""
+
+    def __init__(self, exc):
+        try:
+            serialization.dumps(exc)
+            self.exc = exc
+        except Exception:
+            self.exc = RuntimeError(repr(exc))
+        import traceback
+
+        self.tb = traceback.format_exc()
+
+    def rebuild(self):
+        return self.exc
+
+
+# ---------------------------------------------------------------------------
+# Inventory: seq-keyed result collection (reference fiber/pool.py:644-728)
+# ---------------------------------------------------------------------------
+
+
+class Inventory:
+    def __init__(self):
+        self._lock = threading.Lock()
+        self._cond = threading.Condition(self._lock)
+        self._jobs = {}
+        self._seq = itertools.count()
+
+    def add(self, n):
+        seq = next(self._seq)
+        with self._lock:
+            self._jobs[seq] = {
+                "n": n,
+                "remaining": n,
+                "results": [None] * n,
+                "arrived": [False] * n,
+                "error": None,
+                "order_cursor": 0,
+                "unordered": [],
+            }
+        return seq
+
+    def put(self, seq, base, values, failure):
+        with self._cond:
+            job = self._jobs.get(seq)
+            if job is None:
+                return
+            for offset, value in enumerate(values):
+                index = base + offset
+                if job["arrived"][index]:
+                    continue  # duplicate delivery after a resubmit race
+                job["arrived"][index] = True
+                job["results"][index] = value
+                job["remaining"] -= 1
+                job["unordered"].append((index, value))
+            if failure is not None:
+                index, exc_info = failure
+                if not job["arrived"][index]:
+                    job["arrived"][index] = True
+                    job["remaining"] -= 1
+                    job["error"] = exc_info
+                    job["unordered"].append((index, exc_info))
+                    job["results"][index] = exc_info
+            self._cond.notify_all()
+
+    def fail_all(self, exc):
+        """Abort every outstanding job (pool terminated)."""
+        with self._cond:
+            for job in self._jobs.values():
+                if job["remaining"] > 0:
+                    job["error"] = _ExcInfo(exc)
+                    job["remaining"] = 0
+            self._cond.notify_all()
+
+    def done(self, seq):
+        with self._lock:
+            job = self._jobs.get(seq)
+            return job is None or job["remaining"] == 0
+
+    def get(self, seq, timeout=None):
+        deadline = None if timeout is None else time.monotonic() + timeout
+        with self._cond:
+            job = self._jobs[seq]
+            while job["remaining"] > 0 and job["error"] is None:
+                remaining_t = None
+                if deadline is not None:
+                    remaining_t = deadline - time.monotonic()
+                    if remaining_t <= 0:
+                        raise TimeoutError("pool result timed out")
+                self._cond.wait(remaining_t)
+            if job["error"] is not None:
+                raise job["error"].rebuild()
+            del self._jobs[seq]
+            return job["results"]
+
+    def iget_ordered(self, seq):
+        with self._cond:
+            job = self._jobs[seq]
+        yielded = 0
+        while yielded < job["n"]:
+            with self._cond:
+                while not job["arrived"][yielded] and job["error"] is None:
+                    self._cond.wait()
+                if job["error"] is not None and not job["arrived"][yielded]:
+                    raise job["error"].rebuild()
+                value = job["results"][yielded]
+            if isinstance(value, _ExcInfo):
+                raise value.rebuild()
+            yield value
+            yielded += 1
+        with self._lock:
+            self._jobs.pop(seq, None)
+
+    def iget_unordered(self, seq):
+        with self._cond:
+            job = self._jobs[seq]
+        yielded = 0
+        while yielded < job["n"]:
+            with self._cond:
+                while len(job["unordered"]) <= yielded:
+                    if job["error"] is not None and job["remaining"] == 0:
+                        break
+                    self._cond.wait()
+                if len(job["unordered"]) <= yielded:
+                    raise job["error"].rebuild()
+                _, value = job["unordered"][yielded]
+            if isinstance(value, _ExcInfo):
+                raise value.rebuild()
+            yield value
+            yielded += 1
+        with self._lock:
+            self._jobs.pop(seq, None)
+
+
+class AsyncResult:
+    def __init__(self, pool, seq, n, callback=None, error_callback=None,
+                 single=False):
+        self._pool = pool
+        self._seq = seq
+        self._n = n
+        self._single = single
+        self._callback = callback
+        self._error_callback = error_callback
+
+    def get(self, timeout=None):
+        results = self._pool._inventory.get(self._seq, timeout)
+        first_exc = next(
+            (r for r in results if isinstance(r, _ExcInfo)), None
+        )
+        if first_exc is not None:
+            if self._error_callback:
+                self._error_callback(first_exc.rebuild())
+            raise first_exc.rebuild()
+        out = results[0] if self._single else results
+        if self._callback:
+            self._callback(out)
+        return out
+
+    def wait(self, timeout=None):
+        try:
+            self._pool._inventory.get(self._seq, timeout)
+        except TimeoutError:
+            pass
+
+    def ready(self):
+        return self._pool._inventory.done(self._seq)
+
+    def successful(self):
+        if not self.ready():
+            raise ValueError("result not ready")
+        return True
+
+
+MapResult = AsyncResult
+ApplyResult = AsyncResult
+
+
+# ---------------------------------------------------------------------------
+# Worker side
+# ---------------------------------------------------------------------------
+
+
+def _execute_chunk(func, args, starmap, kwds):
+    values = []
+    for index, arg in enumerate(args):
+        try:
+            if starmap:
+                values.append(func(*arg, **(kwds or {})))
+            else:
+                values.append(func(arg))
+        except Exception as exc:  # noqa: BLE001
+            return values, (index, _ExcInfo(exc))
+    return values, None
+
+
+def _pool_worker_loop(
+    task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix
+):
+    """Worker main loop (reference zpool_worker_core, pool.py:760-825)."""
+    if init_blob is not None:
+        initializer, initargs = serialization.loads(init_blob)
+        initializer(*initargs)
+
+    result_sock = Socket("w", result_addr, bind=False)
+    func_cache = {}
+    tasks_done = 0
+
+    ident = ident_prefix or util.random_name("w")[:24]
+    if resilient:
+        task_sock = Socket("req", task_addr, bind=False, ident=ident)
+    else:
+        task_sock = Socket("r", task_addr, bind=False)
+
+    while True:
+        if resilient:
+            task_sock.send(b"", timeout=-1.0)
+            # Timeout + re-request guards against a discarded request (the
+            # master drops requests it cannot attribute to a live worker).
+            payload = task_sock.recv(timeout=10.0)
+        else:
+            payload = task_sock.recv(timeout=-1.0)
+        if payload is None:
+            continue
+        task = serialization.loads(payload)
+        seq, base, func_blob, args, star, kwds = task
+        if seq == _SENTINEL_SEQ:
+            break
+        key = hash(func_blob)
+        func = func_cache.get(key)
+        if func is None:
+            func = serialization.loads(func_blob)
+            func_cache[key] = func
+
+        values, failure = _execute_chunk(func, args, star, kwds)
+        if failure is not None and resilient:
+            # Resilient pools treat a task exception as a worker fault:
+            # deliver nothing, die, and let the master's pending table
+            # resubmit the whole chunk (tasks must be idempotent —
+            # duplicate deliveries are deduped master-side).
+            raise failure[1].rebuild()
+        result = (seq, base, values, failure, ident)
+        result_sock.send(serialization.dumps(result), timeout=-1.0)
+
+        tasks_done += len(values)
+        if maxtasks is not None and tasks_done >= maxtasks:
+            break
+
+    task_sock.close()
+    result_sock.close()
+
+
+class _WorkerEntry:
+    """Picklable worker target carrying resource metadata for the backend."""
+
+    def __init__(self, kwargs, meta):
+        self._kwargs = kwargs
+        self.__fiber_meta__ = dict(meta or {})
+
+    def __call__(self):
+        _pool_worker_loop(**self._kwargs)
+
+
+# ---------------------------------------------------------------------------
+# Master side
+# ---------------------------------------------------------------------------
+
+
+class ZPool:
+    """Push/pull worker pool with the multiprocessing.Pool API."""
+
+    resilient = False
+
+    def __init__(
+        self,
+        processes=None,
+        initializer=None,
+        initargs=(),
+        maxtasksperchild=None,
+        gpu_per_worker=None,
+        name=None,
+    ):
+        conf = fam_config.get_object()
+        self._processes = processes or os.cpu_count() or 1
+        self._maxtasks = maxtasksperchild
+        self._name = name or util.random_name("fam-pool")
+        self._gpu_per_worker = gpu_per_worker
+        self._meta = {}
+        if gpu_per_worker:
+            self._meta["gpu"] = gpu_per_worker
+
+        self._init_blob = None
+        if initializer is not None:
+            self._init_blob = serialization.dumps_closure(
+                (initializer, tuple(initargs))
+            )
+
+        mode = "rep" if self.resilient else "w"
+        self._task_sock = Socket(mode, self._name + ".task", bind=True)
+        self._result_sock = Socket("r", self._name + ".res", bind=True)
+
+        self._inventory = Inventory()
+        self._taskq = _stdlib_queue.Queue()
+        self._workers = {}  # ident -> Process
+        self._all_idents = set()
+        self._worker_lock = threading.Lock()
+        self._state = "run"  # run -> closing -> terminated
+        self._workers_started = False
+        self._sent = 0
+        self._recv = 0
+        self._max_inflight = conf.max_inflight
+        self._pending = {}  # ident -> {(seq, base): task}  (resilient only)
+
+        self._result_thread = threading.Thread(
+            target=self._result_loop, name="fam-pool-results", daemon=True
+        )
+        self._result_thread.start()
+        self._dispatch_thread = threading.Thread(
+            target=self._dispatch_loop, name="fam-pool-dispatch", daemon=True
+        )
+        self._dispatch_thread.start()
+        self._worker_thread = None
+
+    # -- worker management -------------------------------------------------
+    def _lazy_start_workers(self, func):
+        if self._workers_started:
+            return
+        meta = getattr(func, "__fiber_meta__", None)
+        if meta:
+            if self._meta and self._meta != dict(meta):
+                raise ValueError(
+                    "conflicting resource meta: pool=%r func=%r"
+                    % (self._meta, meta)
+                )
+            self._meta = dict(meta)
+        self._workers_started = True
+        self._worker_thread = threading.Thread(
+            target=self._worker_loop, name="fam-pool-workers", daemon=True
+        )
+        self._worker_thread.start()
+
+    def _spawn_worker(self, index):
+        ident = util.random_name("w")[:24]
+        self._all_idents.add(ident)
+        entry = _WorkerEntry(
+            dict(
+                task_addr=self._name + ".task",
+                result_addr=self._name + ".res",
+                resilient=self.resilient,
+                maxtasks=self._maxtasks,
+                init_blob=self._init_blob,
+                ident_prefix=ident,
+            ),
+            self._meta,
+        )
+        proc = Process(
+            target=entry, name="%s-worker-%d" % (self._name, index)
+        )
+        # Register BEFORE start so the dispatcher can attribute the
+        # worker's very first task request.
+        with self._worker_lock:
+            self._workers[ident] = proc
+        proc.start()
+        return ident, proc
+
+    def _worker_loop(self):
+        """Maintain the worker population; resubmit a dead worker's tasks."""
+        index = itertools.count()
+        while self._state == "run":
+            with self._worker_lock:
+                dead = [
+                    (ident, proc)
+                    for ident, proc in self._workers.items()
+                    if proc.exitcode is not None
+                ]
+                for ident, proc in dead:
+                    del self._workers[ident]
+            for ident, proc in dead:
+                self._on_worker_death(ident, proc)
+            with self._worker_lock:
+                missing = self._processes - len(self._workers)
+            for _ in range(missing):
+                if self._state != "run":
+                    break
+                try:
+                    self._spawn_worker(next(index))
+                except Exception:
+                    if self._state == "run":
+                        util.get_logger().exception("worker spawn failed")
+                    time.sleep(0.5)
+            time.sleep(0.1)
+
+    def _on_worker_death(self, ident, proc):
+        pass  # resilient subclass resubmits
+
+    # -- data plane --------------------------------------------------------
+    def _dispatch_loop(self):
+        """Drain the local task queue into the task ring."""
+        while True:
+            task = self._taskq.get()
+            if task is None:
+                return
+            while (
+                self._sent - self._recv > self._max_inflight
+                and self._state == "run"
+            ):
+                time.sleep(0.005)
+            if self._state == "terminated":
+                continue
+            try:
+                self._send_task(task)
+            except RuntimeError:
+                return  # ring closed
+            self._sent += 1
+
+    def _send_task(self, task):
+        self._task_sock.send(serialization.dumps(task), timeout=-1.0)
+
+    def _result_loop(self):
+        while True:
+            try:
+                payload = self._result_sock._rings["main"].recv(0.2)
+            except RuntimeError:
+                return
+            except Exception:
+                return
+            if self._state == "terminated":
+                return
+            if payload is None:
+                continue
+            seq, base, values, failure, ident = serialization.loads(payload)
+            self._recv += 1
+            self._ack(ident, seq, base)
+            self._inventory.put(seq, base, values, failure)
+
+    def _ack(self, ident, seq, base):
+        pass  # resilient subclass clears pending table
+
+    # -- task submission ---------------------------------------------------
+    def _check_running(self):
+        if self._state != "run":
+            raise ValueError("Pool not running")
+
+    def _submit(self, func, iterable, chunksize, starmap, kwds=None,
+                single=False, callback=None, error_callback=None):
+        self._check_running()
+        self._lazy_start_workers(func)
+        items = list(iterable)
+        n = len(items)
+        seq = self._inventory.add(n)
+        result = AsyncResult(
+            self, seq, n, callback=callback, error_callback=error_callback,
+            single=single,
+        )
+        if n == 0:
+            self._inventory.put(seq, 0, [], None)
+            return result
+        func_blob = serialization.dumps_closure(func)
+        if chunksize is None:
+            chunksize = max(1, min(DEFAULT_CHUNKSIZE, n // 4 or 1))
+        for base in range(0, n, chunksize):
+            chunk = items[base : base + chunksize]
+            self._taskq.put((seq, base, func_blob, chunk, starmap, kwds))
+        return result
+
+    # multiprocessing.Pool API --------------------------------------------
+    def apply(self, func, args=(), kwds=None):
+        return self.apply_async(func, args, kwds).get()
+
+    def apply_async(self, func, args=(), kwds=None, callback=None,
+                    error_callback=None):
+        return self._submit(
+            func, [tuple(args)], 1, True, kwds=kwds or None, single=True,
+            callback=callback, error_callback=error_callback,
+        )
+
+    def map(self, func, iterable, chunksize=None):
+        return self.map_async(func, iterable, chunksize).get()
+
+    def map_async(self, func, iterable, chunksize=None, callback=None,
+                  error_callback=None):
+        return self._submit(
+            func, iterable, chunksize, False, callback=callback,
+            error_callback=error_callback,
+        )
+
+    def starmap(self, func, iterable, chunksize=None):
+        return self.starmap_async(func, iterable, chunksize).get()
+
+    def starmap_async(self, func, iterable, chunksize=None, callback=None,
+                      error_callback=None):
+        return self._submit(
+            func, iterable, chunksize, True, callback=callback,
+            error_callback=error_callback,
+        )
+
+    def imap(self, func, iterable, chunksize=1):
+        result = self._submit(func, iterable, chunksize, False)
+        return self._inventory.iget_ordered(result._seq)
+
+    def imap_unordered(self, func, iterable, chunksize=1):
+        result = self._submit(func, iterable, chunksize, False)
+        return self._inventory.iget_unordered(result._seq)
+
+    # -- lifecycle ---------------------------------------------------------
+    def close(self):
+        if self._state != "run":
+            return
+        self._state = "closing"
+        # One exit sentinel per worker rides the normal task channel (via
+        # the local queue) so it lands after all real tasks.
+        for _ in range(self._processes):
+            self._taskq.put((_SENTINEL_SEQ, 0, b"", [], False, None))
+
+    def terminate(self):
+        if self._state == "terminated":
+            return
+        self._state = "terminated"
+        self._taskq.put(None)
+        with self._worker_lock:
+            for proc in self._workers.values():
+                proc.terminate()
+        self._inventory.fail_all(RuntimeError("pool terminated"))
+
+    def join(self, timeout=30.0):
+        deadline = time.monotonic() + timeout
+        with self._worker_lock:
+            workers = list(self._workers.values())
+        for proc in workers:
+            if proc._popen is None:
+                continue  # registered but never started (shutdown race)
+            proc.join(max(0.1, deadline - time.monotonic()))
+        with self._worker_lock:
+            self._workers.clear()
+        if self._state != "terminated":
+            self._state = "terminated"
+            self._taskq.put(None)
+        # Reap service threads BEFORE closing rings so no thread is blocked
+        # inside a ring call when the mapping goes away / at interpreter
+        # exit (a daemon thread waking into a finalizing interpreter from
+        # C++ aborts the process).
+        self._dispatch_thread.join(timeout=2.0)
+        self._result_thread.join(timeout=2.0)
+        if self._worker_thread is not None:
+            self._worker_thread.join(timeout=2.0)
+        self._task_sock.close()
+        self._result_sock.close()
+        # Workers killed by SIGTERM never ran their atexit cleanup; reap
+        # their per-ident reply rings (resilient mode) master-side.
+        for ident in self._all_idents:
+            try:
+                os.unlink("/dev/shm/%s.task.r.%s" % (self._name, ident))
+            except OSError:
+                pass
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.terminate()
+        self.join()
+
+    def __del__(self):
+        try:
+            if self._state != "terminated":
+                self.terminate()
+        except Exception:
+            pass
+
+
+class ResilientZPool(ZPool):
+    """REQ/REP pool with pending-table resubmission (reference
+    fiber/pool.py:1425-1689)."""
+
+    resilient = True
+
+    def _alive(self, ident):
+        with self._worker_lock:
+            proc = self._workers.get(ident)
+        return proc is not None and proc.exitcode is None
+
+    def _send_task(self, task):
+        # Serve the next *live* worker request, record attribution, reply.
+        # Stale requests from dead workers are discarded (their reply ring
+        # has no reader; replying there would strand the task), and a
+        # reply that fails because the worker died mid-handoff retracts
+        # the task and serves it to the next requester.
+        seq = task[0]
+        payload = serialization.dumps(task)
+        while True:
+            request = self._task_sock.recv_request(timeout=0.2)
+            if request is None:
+                if self._state == "terminated":
+                    raise RuntimeError("pool terminated")
+                continue
+            ident, _ = request
+            if not self._alive(ident):
+                continue
+            if seq != _SENTINEL_SEQ:
+                self._pending.setdefault(ident, {})[(seq, task[1])] = task
+            try:
+                self._task_sock.send_reply(ident, payload)
+            except (RuntimeError, OSError):
+                # Reply ring closed/unlinked: the worker is gone.
+                if seq != _SENTINEL_SEQ:
+                    self._pending.get(ident, {}).pop((seq, task[1]), None)
+                self._task_sock.drop_peer(ident)
+                continue
+            if seq != _SENTINEL_SEQ and not self._alive(ident):
+                # Died between the liveness check and the reply; the reaper
+                # may have already drained its pending table, so pull the
+                # entry back ourselves (duplicate delivery is deduped by
+                # the Inventory's arrived[] bitmap).
+                entry = self._pending.get(ident, {}).pop((seq, task[1]), None)
+                if entry is not None:
+                    self._taskq.put(entry)
+            return
+
+    def _ack(self, ident, seq, base):
+        table = self._pending.get(ident)
+        if table is not None:
+            table.pop((seq, base), None)
+
+    def _on_worker_death(self, ident, proc):
+        """Requeue everything the dead worker had claimed."""
+        table = self._pending.pop(ident, None)
+        self._task_sock.drop_peer(ident)
+        if table:
+            log = util.get_logger()
+            log.warning(
+                "worker %s died with %d pending chunks; resubmitting",
+                ident,
+                len(table),
+            )
+            for task in table.values():
+                self._taskq.put(task)
+
+
+Pool = ResilientZPool

@@ -63,8 +63,8 @@ def _capacity():
     return fam_config.get_object().ring_capacity
 
 
-def _open(name, create, capacity=None):
-    ring = ShmRing(name, create, capacity or _capacity(), 20.0)
+def _open(name, create, capacity=None, open_timeout=20.0):
+    ring = ShmRing(name, create, capacity or _capacity(), open_timeout)
     if create:
         _register_owned(ring)
     return ring
@@ -73,7 +73,7 @@ def _open(name, create, capacity=None):
 class Socket:
     """A mode-string socket bound or connected to an address."""
 
-    def __init__(self, mode, addr=None, bind=False, capacity=None):
+    def __init__(self, mode, addr=None, bind=False, capacity=None, ident=None):
         if mode not in ("r", "w", "rw", "req", "rep"):
             raise ValueError("bad socket mode %r" % mode)
         self.mode = mode
@@ -95,7 +95,7 @@ class Socket:
             self._rings["q"] = _open(self.addr + ".q", True, cap)
             self._reply_cache = {}
         elif mode == "req":
-            self.ident = util.random_name("i")[:24]
+            self.ident = ident or util.random_name("i")[:24]
             self._rings["q"] = _open(self.addr + ".q", False, cap)
             self._rings["r"] = _open(
                 self.addr + ".r." + self.ident, True, cap
@@ -137,18 +137,15 @@ class Socket:
     def send_reply(self, ident, data, timeout=-1.0):
         ring = self._reply_cache.get(ident)
         if ring is None:
-            ring = _open(self.addr + ".r." + ident, False)
+            # Short open timeout: the requester created its reply ring
+            # before its first request, so a missing ring means it died.
+            ring = _open(self.addr + ".r." + ident, False, open_timeout=2.0)
             self._reply_cache[ident] = ring
         return ring.send(data, timeout)
 
     def drop_peer(self, ident):
         """REP side: forget a dead requester's reply ring."""
-        ring = self._reply_cache.pop(ident, None)
-        if ring is not None:
-            try:
-                ring.detach()
-            except Exception:
-                pass
+        self._reply_cache.pop(ident, None)
 
     # -- stats / lifecycle -------------------------------------------------
     @property
@@ -158,20 +155,18 @@ class Socket:
         return sum(r.size for r in self._rings.values())
 
     def close(self):
+        # NOTE: no explicit detach() here — another thread may be blocked
+        # inside ring.recv() on the same mapping; close() wakes it (closed
+        # flag broadcast) and the munmap happens in the ShmRing destructor
+        # once no Python reference holds the ring.
         for ring in self._rings.values():
             try:
                 if ring.is_owner:
                     ring.close()
                     ring.unlink()
-                ring.detach()
             except Exception:
                 pass
         if self.mode == "rep":
-            for ring in self._reply_cache.values():
-                try:
-                    ring.detach()
-                except Exception:
-                    pass
             self._reply_cache.clear()
         self._rings = {}
 
