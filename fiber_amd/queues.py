@@ -118,6 +118,10 @@ class SimpleQueue:
         self._addr = addr or new_address("fam-q")
         self._bound = create
         self._sock = None
+        if create:
+            # Create the segment eagerly so readers/writers in children
+            # can open it regardless of who touches the queue first.
+            self._ensure()
 
     def _ensure(self):
         if self._sock is None:

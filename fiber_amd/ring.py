@@ -1,0 +1,266 @@
+"""Ring: SPMD launcher owning a real collective engine.
+
+The reference Ring (``fiber/experimental/ring.py:58-129``) only does
+rendezvous — members wire up Gloo/Horovod themselves in the initializer.
+Here the Ring is MI355X-native and owns the collectives: every rank is a
+fiber Process pinned to its own GPU, the process group is RCCL over xGMI
+(``torch.distributed`` backend "nccl" == RCCL on ROCm; "gloo" on CPU-only
+nodes), and :class:`RingContext` exposes allreduce / broadcast /
+all_gather / reduce_scatter / barrier plus a bucketed gradient allreduce
+sized for xGMI links (7 p2p links x ~153 GB/s per GPU: a single ring
+collective is per-link bound, so buckets are kept large — default 64 MB —
+to amortize latency while still overlapping with compute).
+"""
+
+import os
+import socket as _socket
+
+from . import serialization, util
+from .process import Process
+
+
+def _free_tcp_port():
+    s = _socket.socket(_socket.AF_INET, _socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+class RingNode:
+    def __init__(self, rank):
+        self.rank = rank
+        self.ip = "127.0.0.1"
+        self.port = None
+        self.connected = False
+
+
+class RingContext:
+    """Per-rank collective context (available inside initializer/func)."""
+
+    def __init__(self, rank, size, backend=None, bucket_mb=64):
+        self.rank = rank
+        self.size = size
+        self._backend = backend
+        self.bucket_bytes = bucket_mb << 20
+        self._initialized = False
+        self._device = None
+
+    # -- setup -------------------------------------------------------------
+    @property
+    def backend(self):
+        if self._backend is None:
+            import torch
+
+            self._backend = "nccl" if torch.cuda.is_available() else "gloo"
+        return self._backend
+
+    @property
+    def device(self):
+        import torch
+
+        if self._device is None:
+            if torch.cuda.is_available():
+                # HIP_VISIBLE_DEVICES pins us to one MI355X => ordinal 0.
+                self._device = torch.device("cuda", 0)
+                torch.cuda.set_device(self._device)
+            else:
+                self._device = torch.device("cpu")
+        return self._device
+
+    def init(self):
+        if self._initialized:
+            return self
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            dist.init_process_group(
+                backend=self.backend,
+                rank=self.rank,
+                world_size=self.size,
+            )
+        self._initialized = True
+        _ = self.device
+        return self
+
+    def shutdown(self):
+        import torch.distributed as dist
+
+        if self._initialized and dist.is_initialized():
+            dist.destroy_process_group()
+        self._initialized = False
+
+    # -- collectives --------------------------------------------------------
+    def allreduce(self, tensor, average=False):
+        import torch.distributed as dist
+
+        self.init()
+        dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+        if average:
+            tensor /= self.size
+        return tensor
+
+    def broadcast(self, tensor, src=0):
+        import torch.distributed as dist
+
+        self.init()
+        dist.broadcast(tensor, src=src)
+        return tensor
+
+    def all_gather(self, tensor):
+        import torch
+        import torch.distributed as dist
+
+        self.init()
+        out = [torch.empty_like(tensor) for _ in range(self.size)]
+        dist.all_gather(out, tensor)
+        return out
+
+    def all_gather_into(self, out, tensor):
+        import torch.distributed as dist
+
+        self.init()
+        dist.all_gather_into_tensor(out, tensor)
+        return out
+
+    def reduce_scatter(self, out, tensor):
+        import torch.distributed as dist
+
+        self.init()
+        dist.reduce_scatter_tensor(out, tensor)
+        return out
+
+    def barrier(self):
+        import torch.distributed as dist
+
+        self.init()
+        dist.barrier()
+
+    def allreduce_grads(self, parameters, average=True):
+        """Bucketed gradient allreduce (replaces the reference example's
+        per-parameter loop, ``examples/ring.py:81-86``, with large flat
+        buckets sized for the xGMI per-link bound)."""
+        import torch
+        import torch.distributed as dist
+
+        self.init()
+        grads = [p.grad for p in parameters if p.grad is not None]
+        if not grads:
+            return
+        bucket, bucket_bytes = [], 0
+        handles = []
+
+        def flush():
+            nonlocal bucket, bucket_bytes
+            if not bucket:
+                return
+            flat = torch._utils._flatten_dense_tensors(bucket)
+            work = dist.all_reduce(flat, async_op=True)
+            handles.append((work, flat, list(bucket)))
+            bucket, bucket_bytes = [], 0
+
+        for grad in grads:
+            bucket.append(grad)
+            bucket_bytes += grad.numel() * grad.element_size()
+            if bucket_bytes >= self.bucket_bytes:
+                flush()
+        flush()
+        for work, flat, bucket_grads in handles:
+            work.wait()
+            if average:
+                flat /= self.size
+            for grad, synced in zip(
+                bucket_grads,
+                torch._utils._unflatten_dense_tensors(flat, bucket_grads),
+            ):
+                grad.copy_(synced)
+
+
+def _ring_target(rank, size, func_blob, init_blob, master_port, backend,
+                 bucket_mb):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ["MASTER_PORT"] = str(master_port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(size)
+    ctx = RingContext(rank, size, backend=backend, bucket_mb=bucket_mb)
+    initializer = serialization.loads(init_blob) if init_blob else None
+    func = serialization.loads(func_blob)
+    try:
+        if initializer is not None:
+            initializer(ctx)
+        func(rank, size)
+    finally:
+        ctx.shutdown()
+
+
+class Ring:
+    """Launch ``size`` copies of ``func(rank, size)``, one MI355X each."""
+
+    def __init__(self, size, func, initializer=None, gpu_per_rank=None,
+                 backend=None, bucket_mb=64):
+        self.size = size
+        self.func = func
+        self.initializer = initializer
+        self.backend = backend
+        self.bucket_mb = bucket_mb
+        if gpu_per_rank is None:
+            import fiber_amd
+
+            gpu_per_rank = 1 if fiber_amd.gpu_count() > 0 else 0
+        self.gpu_per_rank = gpu_per_rank
+        self.members = [RingNode(i) for i in range(size)]
+        self._procs = []
+
+    def run(self, timeout=None):
+        master_port = _free_tcp_port()
+        func_blob = serialization.dumps_closure(self.func)
+        init_blob = (
+            serialization.dumps_closure(self.initializer)
+            if self.initializer
+            else None
+        )
+
+        meta = {"gpu": self.gpu_per_rank} if self.gpu_per_rank else {}
+        for rank in range(self.size):
+            entry = _RingEntry(
+                dict(
+                    rank=rank,
+                    size=self.size,
+                    func_blob=func_blob,
+                    init_blob=init_blob,
+                    master_port=master_port,
+                    backend=self.backend,
+                    bucket_mb=self.bucket_mb,
+                ),
+                meta,
+            )
+            proc = Process(target=entry, name="fam-ring-%d" % rank)
+            proc.start()
+            self._procs.append(proc)
+            self.members[rank].port = master_port
+            self.members[rank].connected = True
+
+        failures = []
+        for proc in self._procs:
+            proc.join(timeout)
+            if proc.exitcode not in (0, None):
+                failures.append((proc.name, proc.exitcode))
+        if failures:
+            for proc in self._procs:
+                proc.terminate()
+            raise RuntimeError("ring ranks failed: %s" % failures)
+
+    def terminate(self):
+        for proc in self._procs:
+            proc.terminate()
+
+
+class _RingEntry:
+    """Picklable per-rank entry carrying GPU meta for the backend."""
+
+    def __init__(self, kwargs, meta):
+        self._kwargs = kwargs
+        self.__fiber_meta__ = dict(meta or {})
+
+    def __call__(self):
+        _ring_target(**self._kwargs)
