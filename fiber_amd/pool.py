@@ -561,8 +561,15 @@ class ZPool:
 
     def join(self, timeout=30.0):
         deadline = time.monotonic() + timeout
+        # Freeze the population first: once the maintainer thread has
+        # exited, no new worker can be spawned behind our back.
+        if self._worker_thread is not None:
+            self._worker_thread.join(timeout=5.0)
         with self._worker_lock:
             workers = list(self._workers.values())
+        if self._state == "terminated":
+            for proc in workers:
+                proc.terminate()  # idempotent; catches post-sweep spawns
         for proc in workers:
             if proc._popen is None:
                 continue  # registered but never started (shutdown race)

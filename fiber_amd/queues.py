@@ -82,6 +82,13 @@ class Connection:
     def addr(self):
         return self._addr
 
+    def __del__(self):
+        if self._bound:
+            try:
+                self.close()
+            except Exception:
+                pass
+
     # -- pickling: re-dial on the far side (never re-create) ---------------
     def __reduce__(self):
         return (Connection, (self._mode, self._addr, False, True))
@@ -167,6 +174,13 @@ class SimpleQueue:
 
     def __reduce__(self):
         return (SimpleQueue, (self._addr, False))
+
+    def __del__(self):
+        if self._bound:
+            try:
+                self.close()
+            except Exception:
+                pass
 
     def __repr__(self):
         return "SimpleQueue(addr=%r)" % self._addr

@@ -1,0 +1,166 @@
+"""Pool tests (parity: reference tests/test_pool.py)."""
+
+import random
+import time
+
+import pytest
+
+import fiber_amd
+from fiber_amd.pool import ResilientZPool, ZPool
+
+
+def _square(x):
+    return x * x
+
+
+def _add(a, b):
+    return a + b
+
+
+def _random_error(x):
+    """Chaos worker (reference random_error_worker, test_pool.py:60-68)."""
+    random.seed()
+    if random.random() < 0.05:
+        raise ValueError("chaos")
+    return x * 2
+
+
+def _slow_identity(x):
+    time.sleep(0.01)
+    return x
+
+
+@pytest.fixture
+def pool():
+    p = ZPool(processes=4)
+    yield p
+    p.terminate()
+    p.join()
+
+
+@pytest.fixture
+def rpool():
+    p = ResilientZPool(processes=4)
+    yield p
+    p.terminate()
+    p.join()
+
+
+class TestZPool:
+    def test_map(self, pool):
+        assert pool.map(_square, range(100)) == [x * x for x in range(100)]
+
+    def test_map_empty(self, pool):
+        assert pool.map(_square, []) == []
+
+    def test_apply(self, pool):
+        assert pool.apply(_add, (2, 3)) == 5
+
+    def test_apply_async(self, pool):
+        assert pool.apply_async(_add, (2, 3)).get(30) == 5
+
+    def test_starmap(self, pool):
+        assert pool.starmap(_add, [(1, 2), (3, 4)]) == [3, 7]
+
+    def test_starmap_async(self, pool):
+        assert pool.starmap_async(_add, [(1, 2), (3, 4)]).get(30) == [3, 7]
+
+    def test_imap_ordered(self, pool):
+        assert list(pool.imap(_square, range(50))) == [
+            x * x for x in range(50)
+        ]
+
+    def test_imap_unordered(self, pool):
+        got = sorted(pool.imap_unordered(_square, range(50)))
+        assert got == sorted(x * x for x in range(50))
+
+    def test_map_async_callback(self, pool):
+        seen = []
+        r = pool.map_async(_square, range(10), callback=seen.append)
+        r.get(30)
+        assert seen and seen[0] == [x * x for x in range(10)]
+
+    def test_large_map(self, pool):
+        n = 5000
+        assert pool.map(_square, range(n)) == [x * x for x in range(n)]
+
+    def test_exception_propagates(self, pool):
+        def boom(x):
+            raise RuntimeError("boom %d" % x)
+
+        with pytest.raises(RuntimeError):
+            pool.map(boom, range(8))
+
+    def test_map_after_close_raises(self):
+        p = ZPool(processes=2)
+        p.map(_square, range(4))
+        p.close()
+        with pytest.raises(ValueError):
+            p.map(_square, range(4))
+        p.join()
+
+    def test_uneven_task_durations_balanced(self, pool):
+        # demand-driven dispatch: a slow task must not serialize the rest
+        t0 = time.monotonic()
+        res = pool.map(_slow_identity, range(80), chunksize=1)
+        elapsed = time.monotonic() - t0
+        assert res == list(range(80))
+        # 80 tasks x 10ms / 4 workers = 0.2s ideal; allow generous slack
+        assert elapsed < 2.0
+
+    def test_context_manager(self):
+        with ZPool(processes=2) as p:
+            assert p.map(_square, range(10)) == [x * x for x in range(10)]
+
+
+class TestResilientZPool:
+    def test_chaos_map_completes(self, rpool):
+        res = rpool.map(_random_error, range(300), chunksize=4)
+        assert res == [x * 2 for x in range(300)]
+
+    def test_chaos_unordered(self, rpool):
+        got = sorted(rpool.imap_unordered(_random_error, range(100),
+                                          chunksize=4))
+        assert got == sorted(x * 2 for x in range(100))
+
+    def test_worker_kill_mid_map_recovers(self, rpool):
+        result = rpool.map_async(_slow_identity, range(200), chunksize=2)
+        time.sleep(0.3)
+        # murder one live worker mid-flight
+        with rpool._worker_lock:
+            victim = next(iter(rpool._workers.values()))
+        victim.kill()
+        assert result.get(60) == list(range(200))
+
+    def test_plain_map(self, rpool):
+        assert rpool.map(_square, range(64)) == [x * x for x in range(64)]
+
+    def test_facade_error_handling_flag(self):
+        p = fiber_amd.Pool(2, error_handling=True)
+        assert isinstance(p, ResilientZPool)
+        p.terminate()
+        p.join()
+        p2 = fiber_amd.Pool(2)
+        assert isinstance(p2, ZPool) and not isinstance(p2, ResilientZPool)
+        p2.terminate()
+        p2.join()
+
+
+class TestPoolInitializer:
+    def test_initializer_runs_in_workers(self):
+        def init(v):
+            import os
+
+            os.environ["FAM_TEST_INIT"] = str(v)
+
+        def read_init(_):
+            import os
+
+            return os.environ.get("FAM_TEST_INIT")
+
+        p = ZPool(processes=2, initializer=init, initargs=(7,))
+        try:
+            assert p.map(read_init, range(4)) == ["7"] * 4
+        finally:
+            p.terminate()
+            p.join()

@@ -1,0 +1,136 @@
+"""Shm-ring transport engine tests."""
+
+import threading
+
+import pytest
+
+from fiber_amd.transport import ShmRing, Socket, new_address
+
+
+class TestShmRing:
+    def test_roundtrip(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 1 << 20, 5.0)
+        try:
+            assert ring.send(b"abc", 1.0)
+            assert ring.recv(1.0) == b"abc"
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_wraparound(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 64 << 10, 5.0)
+        try:
+            msg = b"m" * 5000
+            for _ in range(100):
+                ring.send(msg, 1.0)
+                assert ring.recv(1.0) == msg
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_blocking_backpressure(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 16 << 10, 5.0)
+        try:
+            msg = b"x" * 4000
+            sent = 0
+            while ring.send(msg, 0.0):
+                sent += 1
+            assert 2 <= sent <= 4  # capacity-bounded
+            assert ring.recv(1.0) == msg
+            assert ring.send(msg, 1.0)  # space freed
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_oversize_message_raises(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        try:
+            with pytest.raises(RuntimeError):
+                ring.send(b"y" * (8 << 10), 1.0)
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_closed_ring_raises(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 1 << 20, 5.0)
+        ring.close()
+        with pytest.raises(RuntimeError):
+            ring.recv(1.0)
+        ring.unlink()
+
+    def test_open_missing_times_out(self):
+        with pytest.raises(RuntimeError):
+            ShmRing(new_address("fam-missing"), False, 0, 0.2)
+
+    def test_mpmc_threads(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 20, 5.0)
+        n, nprod = 2000, 4
+        received = []
+        lock = threading.Lock()
+
+        def produce(k):
+            for i in range(n):
+                ring.send(b"%d:%d" % (k, i), -1.0)
+
+        def consume():
+            for _ in range(n):
+                msg = ring.recv(-1.0)
+                with lock:
+                    received.append(msg)
+
+        threads = [
+            threading.Thread(target=produce, args=(k,)) for k in range(nprod)
+        ] + [threading.Thread(target=consume) for _ in range(nprod)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(30)
+        assert len(received) == n * nprod
+        assert len(set(received)) == n * nprod
+        ring.close()
+        ring.unlink()
+
+
+class TestSocket:
+    def test_push_pull(self):
+        addr = new_address()
+        w = Socket("w", addr, bind=True)
+        r = Socket("r", addr, bind=False)
+        w.send(b"task")
+        assert r.recv(1.0) == b"task"
+        w.close()
+        r.close()
+
+    def test_req_rep(self):
+        addr = new_address()
+        rep = Socket("rep", addr, bind=True)
+        req = Socket("req", addr, bind=False)
+        req.send(b"gimme")
+        ident, payload = rep.recv_request(1.0)
+        assert payload == b"gimme"
+        assert ident == req.ident
+        rep.send_reply(ident, b"task-1")
+        assert req.recv(1.0) == b"task-1"
+        req.close()
+        rep.close()
+
+    def test_rw_pair(self):
+        addr = new_address()
+        a = Socket("rw", addr, bind=True)
+        b = Socket("rw", addr, bind=False)
+        a.send(b"ping")
+        assert b.recv(1.0) == b"ping"
+        b.send(b"pong")
+        assert a.recv(1.0) == b"pong"
+        a.close()
+        b.close()
+
+    def test_bad_mode_raises(self):
+        with pytest.raises(ValueError):
+            Socket("xyz", "addr")
