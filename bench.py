@@ -1,0 +1,113 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: OpenAI-ES MLP policy rollouts on MI355X.
+
+BASELINE.json config 3: 2-layer(-hidden) MLP policy on synthetic
+CartPole-dim observations, antithetic-pair perturbations from an on-chip
+Philox noise table, Ring all-reduce over xGMI (RCCL) at N>1.  One "step"
+is one full ES iteration (rollout shard -> all-gather fitness ->
+centered rank -> noise-weighted gradient -> all-reduce -> Adam update).
+
+Run (single GPU):      python bench.py
+Run (N ranks, driver): python -m torch.distributed.run --nnodes=1 \
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Metric: es_rollouts_per_sec — completed env episodes per second, whole
+job (a rollout = one episode of one env instance: pop_total x 64 envs
+per iteration).  Synthetic data, random-init weights (no datasets in
+this environment); weak scaling (per-GPU population fixed).
+"""
+
+import argparse
+import json
+import os
+import time
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=3)
+    parser.add_argument("--pop-per-gpu", type=int, default=4096)
+    parser.add_argument("--horizon", type=int, default=256)
+    args = parser.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    ctx = None
+    if world > 1:
+        from fiber_amd.ring import RingContext
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        ctx = RingContext(rank, world, backend="nccl", device=device)
+        ctx.init()
+
+    from fiber_amd.es import ESConfig, ESEngine
+
+    cfg = ESConfig(pop_per_gpu=args.pop_per_gpu, horizon=args.horizon)
+    engine = ESEngine(cfg, ctx=ctx, device=device)
+
+    for i in range(args.warmup):
+        engine.step(iteration=i)
+
+    if ctx is not None:
+        ctx.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        engine.step(iteration=args.warmup + i)
+    if ctx is not None:
+        ctx.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    if ctx is not None:
+        t = torch.tensor([elapsed], device=device)
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    rollouts_per_step = cfg.pop_per_gpu * world * cfg.envs_per_member
+    value = rollouts_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "es_rollouts_per_sec",
+            "value": value,
+            "unit": "rollouts/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "es-mlp-obs4-h64x64-act2",
+                "global_batch": cfg.pop_per_gpu * world
+                * cfg.envs_per_member,
+                "seq_len": args.horizon,
+                "parallelism": "dp%d" % world,
+                "pop_per_gpu": cfg.pop_per_gpu,
+                "envs_per_member": cfg.envs_per_member,
+                "env_steps_per_sec": value * args.horizon,
+                "sigma": cfg.sigma,
+            },
+        }))
+
+    if ctx is not None:
+        ctx.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,409 @@
+// CDNA4 (gfx950) kernels for the ES engine — the hot path of the rebuild
+// (SURVEY §2e kernel inventory: batched MLP policy forward on MFMA,
+// obs-normalization in LDS, fitness/centered-rank reduction, on-the-fly
+// Philox noise; reference workload: uber/fiber examples/gecco-2020/es.py).
+//
+// Design notes (MI355X-first):
+// * es_rollout_mlp is a PERSISTENT kernel: one workgroup rolls out one
+//   population member's E=64 synthetic-env episodes for the full horizon
+//   T in a single launch — weights, activations and env state never leave
+//   LDS, so the whole ES iteration is one kernel launch per rank plus one
+//   tiny gradient launch (no per-step launch overhead, no HBM traffic for
+//   activations).
+// * Both matmul layers ride MFMA (v_mfma_f32_16x16x32_bf16): layer 1 is
+//   K=32 (obs zero-padded 4->32), layer 2 is K=64.  A-operand = weights
+//   [out][k] row-major in LDS; B-operand = activations [env][k] row-major
+//   ("transposed" storage so each lane's 8 k-consecutive bf16 are one
+//   ds_read_b128).
+// * Perturbations are regenerated from Philox counters (philox.h), never
+//   stored: member 2k applies +sigma*eps_k, member 2k+1 applies
+//   -sigma*eps_k (antithetic pairs).
+//
+// Fragment layouts for v_mfma_f32_16x16x32_bf16 (A 16x32, B 32x16,
+// C/D 16x16), per the CDNA4 guide + hardware verification test
+// (tests/gpu/test_ops.py::test_mfma_gemm_against_torch):
+//   A: lane l holds A[row=l&15][k=(l>>4)*8 + j], j=0..7
+//   B: lane l holds B[k=(l>>4)*8 + j][col=l&15]
+//   C/D: lane l, reg r holds D[row=(l>>4)*4 + r][col=l&15]
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include "philox.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FAM_TAG_NOISE 0x45530001u
+#define FAM_TAG_ENV 0x45530002u
+
+// ---------------------------------------------------------------------------
+// Policy geometry (compile-time; the flagship BASELINE config).
+// ---------------------------------------------------------------------------
+#define OBS 4
+#define KPAD 32  // obs padded to one MFMA K-tile
+#define HID 64
+#define ACT 2
+#define ENVS 64  // env instances per member (common random numbers)
+
+// flat theta layout offsets
+#define OFF_W1 0
+#define OFF_B1 (HID * OBS)                  // 256
+#define OFF_W2 (OFF_B1 + HID)               // 320
+#define OFF_B2 (OFF_W2 + HID * HID)         // 4416
+#define OFF_W3 (OFF_B2 + HID)               // 4480
+#define OFF_B3 (OFF_W3 + ACT * HID)         // 4608
+#define NPARAMS (OFF_B3 + ACT)              // 4610
+
+#define S1 40  // LDS stride (bf16) for K=32 operands
+#define S2 72  // LDS stride (bf16) for K=64 operands
+
+// One wave computes a 16-row strip of C[64][64] = A[64][K] x B[K][64],
+// reading A as [row][k] (stride AS) and B as [col][k] (stride BS), then
+// applies bias+tanh and stores C transposed into out[col][row] (stride OS)
+// so the output is ready as the next layer's B-operand.
+template <int K, int AS, int BS, int OS>
+__device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
+                                       const __hip_bfloat16* __restrict__ B,
+                                       const float* __restrict__ bias,
+                                       __hip_bfloat16* __restrict__ out,
+                                       int wave, int lane) {
+  const int r0 = wave * 16;
+  const int arow = r0 + (lane & 15);
+  const int kgrp = lane >> 4;  // 0..3
+
+  bf16x8 afrag[K / 32];
+#pragma unroll
+  for (int kk = 0; kk < K / 32; ++kk) {
+    afrag[kk] = *reinterpret_cast<const bf16x8*>(
+        &A[arow * AS + kk * 32 + kgrp * 8]);
+  }
+
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int bcol = ct * 16 + (lane & 15);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < K / 32; ++kk) {
+      bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &B[bcol * BS + kk * 32 + kgrp * 8]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bfrag, acc,
+                                                    0, 0, 0);
+    }
+    // D: row = r0 + kgrp*4 + ri, col = bcol.  The 4 rows are consecutive:
+    // pack one 8-byte store into out[col][row..row+3].
+    const int drow = r0 + kgrp * 4;
+    union {
+      __hip_bfloat16 h[4];
+      unsigned long long u;
+    } pk;
+#pragma unroll
+    for (int ri = 0; ri < 4; ++ri) {
+      float v = tanhf(acc[ri] + bias[drow + ri]);
+      pk.h[ri] = __float2bfloat16(v);
+    }
+    *reinterpret_cast<unsigned long long*>(&out[bcol * OS + drow]) = pk.u;
+  }
+}
+
+// Decompose flat theta index j -> LDS slot write.
+struct PolicyLds {
+  alignas(16) __hip_bfloat16 w1[HID][S1];
+  alignas(16) __hip_bfloat16 w2[HID][S2];
+  alignas(16) __hip_bfloat16 w3[ACT][HID];
+  float b1[HID];
+  float b2[HID];
+  float b3[ACT];
+};
+
+__device__ inline void store_param(PolicyLds* p, int j, float v) {
+  if (j < OFF_B1) {
+    p->w1[j >> 2][j & 3] = __float2bfloat16(v);
+  } else if (j < OFF_W2) {
+    p->b1[j - OFF_B1] = v;
+  } else if (j < OFF_B2) {
+    int jj = j - OFF_W2;
+    p->w2[jj >> 6][jj & 63] = __float2bfloat16(v);
+  } else if (j < OFF_W3) {
+    p->b2[j - OFF_B2] = v;
+  } else if (j < OFF_B3) {
+    int jj = j - OFF_W3;
+    p->w3[jj >> 6][jj & 63] = __float2bfloat16(v);
+  } else if (j < NPARAMS) {
+    p->b3[j - OFF_B3] = v;
+  }
+}
+
+// Shared epilogue/prologue state for one member rollout.
+struct RolloutLds {
+  PolicyLds pol;
+  alignas(16) __hip_bfloat16 xb[ENVS][S1];  // normalized obs, layer-1 B
+  alignas(16) __hip_bfloat16 h1[ENVS][S2];  // B-operand layer 2
+  alignas(16) __hip_bfloat16 h2[ENVS][S2];  // input to logits
+  float S[ENVS][OBS];                       // env state
+  float racc[ENVS];                         // per-env return
+  float ostat[2 * OBS + 1];                 // sum, sumsq, count
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
+               uint32_t iter, int horizon, int member_offset,
+               const float* __restrict__ obs_mu,
+               const float* __restrict__ obs_nu,
+               const float* __restrict__ env_A,   // [OBS][OBS]
+               const float* __restrict__ env_B,   // [OBS]
+               float* __restrict__ fitness,       // [pop_shard]
+               float* __restrict__ obs_stat_out)  // [2*OBS+1]
+{
+  __shared__ RolloutLds L;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int member = member_offset + blockIdx.x;
+  const uint32_t pair = (uint32_t)(member >> 1);
+  const float sgn = (member & 1) ? -sigma : sigma;
+
+  // ---- fill perturbed policy into LDS ---------------------------------
+  for (int jb = tid; jb * 4 < NPARAMS; jb += blockDim.x) {
+    float z[4];
+    fam_normal4(seed, iter, pair, (uint32_t)jb, FAM_TAG_NOISE, 0u, z);
+    const int j0 = jb * 4;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int j = j0 + u;
+      if (j < NPARAMS) store_param(&L.pol, j, theta[j] + sgn * z[u]);
+    }
+  }
+  // zero the K-padding of W1 (A-side zeros make pad products zero) and xb
+  for (int idx = tid; idx < HID * (KPAD - OBS); idx += blockDim.x) {
+    L.pol.w1[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+        __float2bfloat16(0.f);
+  }
+  for (int idx = tid; idx < ENVS * (KPAD - OBS); idx += blockDim.x) {
+    L.xb[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+        __float2bfloat16(0.f);
+  }
+  // init env state (member-independent: common random numbers), zero accs
+  if (tid < ENVS) {
+    float z[4];
+    fam_normal4(seed, iter, (uint32_t)tid, 0u, FAM_TAG_ENV, 0u, z);
+#pragma unroll
+    for (int d = 0; d < OBS; ++d) L.S[tid][d] = 0.3f * z[d];
+    L.racc[tid] = 0.f;
+  }
+  if (tid < 2 * OBS + 1) L.ostat[tid] = 0.f;
+  __syncthreads();
+
+  // per-env obs-stat partials live in the env-owner thread's registers
+  float psum[OBS] = {0.f};
+  float psq[OBS] = {0.f};
+  float mu[OBS], rstd[OBS];
+#pragma unroll
+  for (int d = 0; d < OBS; ++d) {
+    mu[d] = obs_mu[d];
+    rstd[d] = rsqrtf(obs_nu[d] + 1e-4f);
+  }
+  float eA[OBS][OBS], eB[OBS];
+#pragma unroll
+  for (int d = 0; d < OBS; ++d) {
+#pragma unroll
+    for (int e = 0; e < OBS; ++e) eA[d][e] = env_A[d * OBS + e];
+    eB[d] = env_B[d];
+  }
+
+  for (int t = 0; t < horizon; ++t) {
+    // ---- phase A: normalize obs into xb (LDS), accumulate raw stats ---
+    if (tid < ENVS) {
+#pragma unroll
+      for (int d = 0; d < OBS; ++d) {
+        float s = L.S[tid][d];
+        psum[d] += s;
+        psq[d] += s * s;
+        float x = (s - mu[d]) * rstd[d];
+        x = fminf(5.f, fmaxf(-5.f, x));
+        L.xb[tid][d] = __float2bfloat16(x);
+      }
+    }
+    __syncthreads();
+    // ---- phase B: h1 = tanh(W1 x + b1)  (MFMA, K=32) ------------------
+    mfma_strip_tanh<KPAD, S1, S1, S2>(&L.pol.w1[0][0], &L.xb[0][0],
+                                      L.pol.b1, &L.h1[0][0], wave, lane);
+    __syncthreads();
+    // ---- phase C: h2 = tanh(W2 h1 + b2) (MFMA, K=64) ------------------
+    mfma_strip_tanh<HID, S2, S2, S2>(&L.pol.w2[0][0], &L.h1[0][0],
+                                     L.pol.b2, &L.h2[0][0], wave, lane);
+    __syncthreads();
+    // ---- phase D: logits + argmax action + synthetic env step ---------
+    if (tid < ENVS) {
+      float l0 = L.pol.b3[0], l1 = L.pol.b3[1];
+      for (int h = 0; h < HID; ++h) {
+        float hv = __bfloat162float(L.h2[tid][h]);
+        l0 += __bfloat162float(L.pol.w3[0][h]) * hv;
+        l1 += __bfloat162float(L.pol.w3[1][h]) * hv;
+      }
+      const float asign = (l1 > l0) ? 1.f : -1.f;
+      float snew[OBS];
+      float sq = 0.f;
+#pragma unroll
+      for (int d = 0; d < OBS; ++d) {
+        float drive = 0.f;
+#pragma unroll
+        for (int e = 0; e < OBS; ++e) drive += eA[d][e] * L.S[tid][e];
+        snew[d] = 0.97f * L.S[tid][d] + 0.08f * tanhf(drive) +
+                  0.05f * eB[d] * asign;
+        sq += snew[d] * snew[d];
+      }
+      L.racc[tid] += 1.f - 0.1f * sq;
+#pragma unroll
+      for (int d = 0; d < OBS; ++d) L.S[tid][d] = snew[d];
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: fitness + obs-stat reduction --------------------------
+  if (tid < ENVS) {
+#pragma unroll
+    for (int d = 0; d < OBS; ++d) {
+      atomicAdd(&L.ostat[d], psum[d]);
+      atomicAdd(&L.ostat[OBS + d], psq[d]);
+    }
+  }
+  __syncthreads();
+  if (wave == 0) {
+    float r = L.racc[lane];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) r += __shfl_down(r, off, 64);
+    if (lane == 0) fitness[blockIdx.x] = r / (float)ENVS;
+  }
+  if (tid < 2 * OBS) atomicAdd(&obs_stat_out[tid], L.ostat[tid]);
+  if (tid == 0)
+    atomicAdd(&obs_stat_out[2 * OBS], (float)(ENVS * horizon));
+}
+
+// ---------------------------------------------------------------------------
+// Gradient: g[j] = sum_pairs w_k * eps_k[j]  (eps regenerated via Philox).
+// grid.x tiles the parameter blocks, grid.y tiles the local pair range.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void es_grad(const float* __restrict__ wpair,
+                                   int pair_begin, int pair_end,
+                                   int pairs_per_chunk, uint32_t seed,
+                                   uint32_t iter,
+                                   float* __restrict__ grad) {
+  const int jb = blockIdx.x * blockDim.x + threadIdx.x;
+  if (jb * 4 >= NPARAMS) return;
+  const int chunk_begin = pair_begin + blockIdx.y * pairs_per_chunk;
+  const int chunk_end = min(chunk_begin + pairs_per_chunk, pair_end);
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int pair = chunk_begin; pair < chunk_end; ++pair) {
+    const float w = wpair[pair];
+    if (w == 0.f) continue;
+    float z[4];
+    fam_normal4(seed, iter, (uint32_t)pair, (uint32_t)jb, FAM_TAG_NOISE, 0u,
+                z);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc[u] += w * z[u];
+  }
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    const int j = jb * 4 + u;
+    if (j < NPARAMS && (acc[u] != 0.f)) atomicAdd(&grad[j], acc[u]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Centered rank: out[i] = rank(f_i)/(n-1) - 0.5 with deterministic
+// index-order tie-break.  O(n^2/threads) — n is the population (<=64k).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void centered_rank(const float* __restrict__ f, int n,
+                                         float* __restrict__ out) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const float fi = f[i];
+  int rank = 0;
+  for (int j = 0; j < n; ++j) {
+    const float fj = f[j];
+    rank += (fj < fi) || (fj == fi && j < i);
+  }
+  out[i] = (float)rank / (float)(n - 1) - 0.5f;
+}
+
+// ---------------------------------------------------------------------------
+// Standalone batched policy forward (numerics target + serving op):
+// logits[b][ACT] for X[b][OBS], unperturbed theta.  One workgroup per
+// 64-row batch tile; same MFMA path as the rollout kernel.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+mlp_policy_forward(const float* __restrict__ theta,
+                   const float* __restrict__ X, int batch,
+                   float* __restrict__ logits) {
+  __shared__ PolicyLds pol;
+  __shared__ alignas(16) __hip_bfloat16 xb[ENVS][S1];
+  __shared__ alignas(16) __hip_bfloat16 h1[ENVS][S2];
+  __shared__ alignas(16) __hip_bfloat16 h2[ENVS][S2];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * ENVS;
+
+  for (int j = tid; j < NPARAMS; j += blockDim.x) store_param(&pol, j, theta[j]);
+  for (int idx = tid; idx < HID * (KPAD - OBS); idx += blockDim.x)
+    pol.w1[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+        __float2bfloat16(0.f);
+  for (int idx = tid; idx < ENVS * KPAD; idx += blockDim.x) {
+    int e = idx / KPAD, d = idx % KPAD;
+    float v = (d < OBS && row0 + e < batch) ? X[(row0 + e) * OBS + d] : 0.f;
+    xb[e][d] = __float2bfloat16(v);
+  }
+  __syncthreads();
+  mfma_strip_tanh<KPAD, S1, S1, S2>(&pol.w1[0][0], &xb[0][0], pol.b1,
+                                    &h1[0][0], wave, lane);
+  __syncthreads();
+  mfma_strip_tanh<HID, S2, S2, S2>(&pol.w2[0][0], &h1[0][0], pol.b2,
+                                   &h2[0][0], wave, lane);
+  __syncthreads();
+  if (tid < ENVS && row0 + tid < batch) {
+#pragma unroll
+    for (int a = 0; a < ACT; ++a) {
+      float l = pol.b3[a];
+      for (int h = 0; h < HID; ++h)
+        l += __bfloat162float(pol.w3[a][h]) *
+             __bfloat162float(h2[tid][h]);
+      logits[(row0 + tid) * ACT + a] = l;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Raw MFMA refcheck kernel: C[64][64] = tanh-free A[64][64] x B[64][64]
+// through the exact strip path used above (bias=0, output de-transposed
+// host-side).  Lets the GPU test verify the fragment-layout assumptions
+// with random asymmetric inputs (guide ERRATA #3: symmetric inputs can
+// hide a transposed C-write).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+mfma_gemm64_probe(const float* __restrict__ A64,
+                  const float* __restrict__ B64,
+                  float* __restrict__ C64T) {
+  __shared__ alignas(16) __hip_bfloat16 a[HID][S2];
+  __shared__ alignas(16) __hip_bfloat16 b[HID][S2];   // stored [col][k]
+  __shared__ alignas(16) __hip_bfloat16 c[HID][S2];   // written [col][row]
+  __shared__ float zero_bias[HID];
+  const int tid = threadIdx.x;
+  for (int idx = tid; idx < HID * HID; idx += blockDim.x) {
+    int r = idx >> 6, k = idx & 63;
+    a[r][k] = __float2bfloat16(A64[r * HID + k]);
+    // B stored transposed: b[col][k] = B[k][col]
+    b[r][k] = __float2bfloat16(B64[k * HID + r]);
+  }
+  if (tid < HID) zero_bias[tid] = 0.f;
+  __syncthreads();
+  // tanh is applied by the strip helper; host compares against
+  // tanh(A@B) to keep one code path.
+  mfma_strip_tanh<HID, S2, S2, S2>(&a[0][0], &b[0][0], zero_bias, &c[0][0],
+                                   tid >> 6, tid & 63);
+  __syncthreads();
+  for (int idx = tid; idx < HID * HID; idx += blockDim.x) {
+    int col = idx >> 6, row = idx & 63;
+    C64T[col * HID + row] = __bfloat162float(c[col][row]);
+  }
+}

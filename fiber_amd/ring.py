@@ -38,13 +38,17 @@ class RingNode:
 class RingContext:
     """Per-rank collective context (available inside initializer/func)."""
 
-    def __init__(self, rank, size, backend=None, bucket_mb=64):
+    def __init__(self, rank, size, backend=None, bucket_mb=64, device=None):
         self.rank = rank
         self.size = size
         self._backend = backend
         self.bucket_bytes = bucket_mb << 20
         self._initialized = False
-        self._device = None
+        self._device = device
+        if device is not None and device.type == "cuda":
+            import torch
+
+            torch.cuda.set_device(device)
 
     # -- setup -------------------------------------------------------------
     @property

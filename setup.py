@@ -1,0 +1,89 @@
+"""In-tree builds for the fiber_amd native extensions.
+
+  python setup.py build_ext --inplace
+
+Builds:
+  fiber_amd._transport — C++ shm-ring engine (host-only, g++)
+  fiber_amd._ops       — CDNA4 HIP kernels (hipcc --offload-arch=gfx950)
+
+The HIP extension is compiled with hipcc directly (no torch headers), so
+the whole build cross-compiles on a GPU-less box in seconds.
+"""
+
+import os
+import subprocess
+import sys
+
+import pybind11
+from setuptools import setup
+from setuptools.command.build_ext import build_ext as _build_ext
+from setuptools.extension import Extension
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+
+
+def ext_suffix():
+    import sysconfig
+
+    return sysconfig.get_config_var("EXT_SUFFIX")
+
+
+class build_ext(_build_ext):
+    def run(self):
+        super().run()
+        self.build_hip_ops()
+
+    def build_hip_ops(self):
+        out = os.path.join(ROOT, "fiber_amd", "_ops" + ext_suffix())
+        srcs = [
+            os.path.join(ROOT, "fiber_amd", "csrc", "ops", "es_kernels.hip"),
+            os.path.join(ROOT, "fiber_amd", "csrc", "ops", "bindings.cpp"),
+        ]
+        hdrs = [
+            os.path.join(ROOT, "fiber_amd", "csrc", "ops", "philox.h"),
+        ]
+        if os.path.exists(out):
+            newest_src = max(os.path.getmtime(s) for s in srcs + hdrs)
+            if os.path.getmtime(out) >= newest_src:
+                print("fiber_amd._ops up to date")
+                return
+        cmd = [
+            HIPCC,
+            "--offload-arch=" + GFX_ARCH,
+            "-O3",
+            "-std=c++17",
+            "-fPIC",
+            "-shared",
+            "-I",
+            pybind11.get_include(),
+            "-I",
+            os.path.join(sys.prefix, "include", "python%d.%d" % sys.version_info[:2]),
+            "-I",
+            "/usr/include/python%d.%d" % sys.version_info[:2],
+        ] + srcs + ["-o", out]
+        print(" ".join(cmd))
+        subprocess.check_call(cmd)
+
+
+setup(
+    name="fiber_amd",
+    version="0.1.0",
+    packages=[
+        "fiber_amd",
+        "fiber_amd.backends",
+        "fiber_amd.ops",
+        "fiber_amd.es",
+    ],
+    ext_modules=[
+        Extension(
+            "fiber_amd._transport",
+            sources=["fiber_amd/csrc/transport.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O2", "-std=c++17"],
+            libraries=["pthread", "rt"],
+        )
+    ],
+    cmdclass={"build_ext": build_ext},
+)

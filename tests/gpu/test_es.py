@@ -1,0 +1,93 @@
+"""GPU tests for the ES engine: rollout numerics vs fp32 reference,
+determinism, and end-to-end learning."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available(), reason="needs MI355X"
+)
+
+if torch.cuda.is_available():
+    from fiber_amd import ops
+    from fiber_amd.es import ESConfig, ESEngine, engine as es_engine
+
+
+@requires_gpu
+class TestRolloutNumerics:
+    def _setup(self):
+        device = torch.device("cuda")
+        theta = es_engine.init_theta(42, device)
+        env_A, env_B = es_engine.make_env_params(42, device)
+        obs_mu = torch.zeros(4, device=device)
+        obs_nu = torch.ones(4, device=device)
+        return device, theta, env_A, env_B, obs_mu, obs_nu
+
+    def test_fitness_vs_fp32_reference(self):
+        device, theta, env_A, env_B, obs_mu, obs_nu = self._setup()
+        horizon, shard = 8, 4
+        fit, _stat = ops.es_rollout_mlp(
+            theta, 0.05, 1234, 0, horizon, 0, shard, obs_mu, obs_nu,
+            env_A, env_B,
+        )
+        torch.cuda.synchronize()
+        ref = es_engine.rollout_reference(
+            theta, 0.05, 1234, 0, horizon, list(range(shard)), obs_mu,
+            obs_nu, env_A, env_B,
+        )
+        err = (fit.cpu() - ref).abs().max().item()
+        assert err < 5e-2, (fit.cpu(), ref)
+
+    def test_rollout_deterministic(self):
+        device, theta, env_A, env_B, obs_mu, obs_nu = self._setup()
+        f1, s1 = ops.es_rollout_mlp(theta, 0.05, 7, 3, 32, 0, 64, obs_mu,
+                                    obs_nu, env_A, env_B)
+        f2, s2 = ops.es_rollout_mlp(theta, 0.05, 7, 3, 32, 0, 64, obs_mu,
+                                    obs_nu, env_A, env_B)
+        torch.cuda.synchronize()
+        assert torch.equal(f1, f2)
+        assert torch.allclose(s1, s2, rtol=1e-6)
+
+    def test_antithetic_pairs_differ(self):
+        device, theta, env_A, env_B, obs_mu, obs_nu = self._setup()
+        fit, _ = ops.es_rollout_mlp(theta, 0.2, 99, 0, 32, 0, 8, obs_mu,
+                                    obs_nu, env_A, env_B)
+        torch.cuda.synchronize()
+        fit = fit.cpu()
+        # +eps and -eps members must generally produce different fitness
+        assert (fit[0::2] - fit[1::2]).abs().max().item() > 0
+
+    def test_obs_stat_counts(self):
+        device, theta, env_A, env_B, obs_mu, obs_nu = self._setup()
+        horizon, shard = 16, 32
+        _fit, stat = ops.es_rollout_mlp(theta, 0.05, 1, 0, horizon, 0,
+                                        shard, obs_mu, obs_nu, env_A,
+                                        env_B)
+        torch.cuda.synchronize()
+        # count accumulates ENVS*horizon per member
+        assert stat[-1].item() == shard * ops.ENVS_PER_MEMBER * horizon
+
+
+@requires_gpu
+class TestESEngine:
+    def test_single_gpu_step(self):
+        cfg = ESConfig(pop_per_gpu=128, horizon=32)
+        eng = ESEngine(cfg, ctx=None, device=torch.device("cuda", 0))
+        s1 = eng.step()
+        s2 = eng.step()
+        torch.cuda.synchronize()
+        assert s1["grad_norm"] > 0
+        assert s1["rollouts"] == 128 * cfg.envs_per_member
+        assert s2["fitness_mean"] == s2["fitness_mean"]  # not NaN
+
+    def test_learning_improves_fitness(self):
+        torch.manual_seed(0)
+        cfg = ESConfig(pop_per_gpu=512, horizon=64, lr=0.05, sigma=0.1)
+        eng = ESEngine(cfg, ctx=None, device=torch.device("cuda", 0))
+        history = [eng.step()["fitness_mean"] for _ in range(30)]
+        torch.cuda.synchronize()
+        early = sum(history[:5]) / 5
+        late = sum(history[-5:]) / 5
+        assert late > early, history
