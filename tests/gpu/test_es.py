@@ -48,7 +48,9 @@ class TestRolloutNumerics:
                                     obs_nu, env_A, env_B)
         torch.cuda.synchronize()
         assert torch.equal(f1, f2)
-        assert torch.allclose(s1, s2, rtol=1e-6)
+        # obs stats are accumulated with float atomics across workgroups;
+        # ordering is nondeterministic, so low bits may differ.
+        assert torch.allclose(s1, s2, rtol=1e-4)
 
     def test_antithetic_pairs_differ(self):
         device, theta, env_A, env_B, obs_mu, obs_nu = self._setup()
