@@ -34,6 +34,17 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// tanh via the hardware transcendental unit (v_exp_f32):
+// tanh(x) = 1 - 2/(e^{2x}+1).  ~6 VALU slots vs ~40 for libm tanhf —
+// the rollout kernel evaluates 8192 activations per member-step, and the
+// PMC profile (profiles/r01_rollout_pmc.md) showed the libm version made
+// the kernel VALU-bound at 3.6% MfmaUtil.  exp overflow saturates
+// correctly (inf -> tanh=1).
+__device__ inline float fast_tanh(float x) {
+  float e = __expf(2.0f * x);
+  return 1.0f - 2.0f / (e + 1.0f);
+}
+
 #define FAM_TAG_NOISE 0x45530001u
 #define FAM_TAG_ENV 0x45530002u
 
@@ -99,7 +110,7 @@ __device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
     } pk;
 #pragma unroll
     for (int ri = 0; ri < 4; ++ri) {
-      float v = tanhf(acc[ri] + bias[drow + ri]);
+      float v = fast_tanh(acc[ri] + bias[drow + ri]);
       pk.h[ri] = __float2bfloat16(v);
     }
     *reinterpret_cast<unsigned long long*>(&out[bcol * OS + drow]) = pk.u;
@@ -140,9 +151,10 @@ struct RolloutLds {
   alignas(16) __hip_bfloat16 xb[ENVS][S1];  // normalized obs, layer-1 B
   alignas(16) __hip_bfloat16 h1[ENVS][S2];  // B-operand layer 2
   alignas(16) __hip_bfloat16 h2[ENVS][S2];  // input to logits
-  float S[ENVS][OBS];                       // env state
-  float racc[ENVS];                         // per-env return
-  float ostat[2 * OBS + 1];                 // sum, sumsq, count
+  float S[2][ENVS][OBS];                    // env state (double-buffered)
+  float part[OBS][ENVS][2];  // logits partials / squared-state partials
+  float racc[ENVS];          // per-env return
+  float ostat[2 * OBS + 1];  // sum, sumsq, count
 };
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -188,40 +200,40 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     float z[4];
     fam_normal4(seed, iter, (uint32_t)tid, 0u, FAM_TAG_ENV, 0u, z);
 #pragma unroll
-    for (int d = 0; d < OBS; ++d) L.S[tid][d] = 0.3f * z[d];
+    for (int d = 0; d < OBS; ++d) L.S[0][tid][d] = 0.3f * z[d];
     L.racc[tid] = 0.f;
   }
   if (tid < 2 * OBS + 1) L.ostat[tid] = 0.f;
   __syncthreads();
 
-  // per-env obs-stat partials live in the env-owner thread's registers
-  float psum[OBS] = {0.f};
-  float psq[OBS] = {0.f};
-  float mu[OBS], rstd[OBS];
+  // Every phase below uses all 256 threads as (env = tid&63, d = tid>>6)
+  // — OBS==4 matches the 4 waves exactly.  Per-thread scalar obs-stat
+  // partials; reward partials go through L.part to stay deterministic
+  // (no float atomics on the fitness path).
+  const int env = tid & 63;
+  const int dd = tid >> 6;  // 0..3
+  float psum = 0.f, psq = 0.f;
+  const float mu_d = obs_mu[dd];
+  const float rstd_d = rsqrtf(obs_nu[dd] + 1e-4f);
+  float eA_d[OBS];
 #pragma unroll
-  for (int d = 0; d < OBS; ++d) {
-    mu[d] = obs_mu[d];
-    rstd[d] = rsqrtf(obs_nu[d] + 1e-4f);
-  }
-  float eA[OBS][OBS], eB[OBS];
-#pragma unroll
-  for (int d = 0; d < OBS; ++d) {
-#pragma unroll
-    for (int e = 0; e < OBS; ++e) eA[d][e] = env_A[d * OBS + e];
-    eB[d] = env_B[d];
-  }
+  for (int e = 0; e < OBS; ++e) eA_d[e] = env_A[dd * OBS + e];
+  const float eB_d = env_B[dd];
 
   for (int t = 0; t < horizon; ++t) {
-    // ---- phase A: normalize obs into xb (LDS), accumulate raw stats ---
-    if (tid < ENVS) {
-#pragma unroll
-      for (int d = 0; d < OBS; ++d) {
-        float s = L.S[tid][d];
-        psum[d] += s;
-        psq[d] += s * s;
-        float x = (s - mu[d]) * rstd[d];
-        x = fminf(5.f, fmaxf(-5.f, x));
-        L.xb[tid][d] = __float2bfloat16(x);
+    const int cur = t & 1;
+    // ---- phase A: normalize obs -> xb; stats; fold reward partials ----
+    {
+      const float s = L.S[cur][env][dd];
+      psum += s;
+      psq += s * s;
+      float x = (s - mu_d) * rstd_d;
+      x = fminf(5.f, fmaxf(-5.f, x));
+      L.xb[env][dd] = __float2bfloat16(x);
+      if (dd == 0 && t > 0) {
+        // reward for step t-1 (squared-state partials staged in L.part)
+        L.racc[env] += 1.f - 0.1f * (L.part[0][env][0] + L.part[1][env][0] +
+                                     L.part[2][env][0] + L.part[3][env][0]);
       }
     }
     __syncthreads();
@@ -233,41 +245,46 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     mfma_strip_tanh<HID, S2, S2, S2>(&L.pol.w2[0][0], &L.h1[0][0],
                                      L.pol.b2, &L.h2[0][0], wave, lane);
     __syncthreads();
-    // ---- phase D: logits + argmax action + synthetic env step ---------
-    if (tid < ENVS) {
-      float l0 = L.pol.b3[0], l1 = L.pol.b3[1];
-      for (int h = 0; h < HID; ++h) {
-        float hv = __bfloat162float(L.h2[tid][h]);
-        l0 += __bfloat162float(L.pol.w3[0][h]) * hv;
-        l1 += __bfloat162float(L.pol.w3[1][h]) * hv;
+    // ---- phase D1: logits partials (thread (env, quarter)) ------------
+    {
+      float p0 = 0.f, p1 = 0.f;
+#pragma unroll
+      for (int hh = 0; hh < HID / 4; ++hh) {
+        const int h = dd * (HID / 4) + hh;
+        const float hv = __bfloat162float(L.h2[env][h]);
+        p0 += __bfloat162float(L.pol.w3[0][h]) * hv;
+        p1 += __bfloat162float(L.pol.w3[1][h]) * hv;
       }
+      L.part[dd][env][0] = p0;
+      L.part[dd][env][1] = p1;
+    }
+    __syncthreads();
+    // ---- phase D2: action + env step (thread (env, d)) ----------------
+    {
+      const float l0 = L.pol.b3[0] + L.part[0][env][0] + L.part[1][env][0] +
+                       L.part[2][env][0] + L.part[3][env][0];
+      const float l1 = L.pol.b3[1] + L.part[0][env][1] + L.part[1][env][1] +
+                       L.part[2][env][1] + L.part[3][env][1];
       const float asign = (l1 > l0) ? 1.f : -1.f;
-      float snew[OBS];
-      float sq = 0.f;
+      float drive = 0.f;
 #pragma unroll
-      for (int d = 0; d < OBS; ++d) {
-        float drive = 0.f;
-#pragma unroll
-        for (int e = 0; e < OBS; ++e) drive += eA[d][e] * L.S[tid][e];
-        snew[d] = 0.97f * L.S[tid][d] + 0.08f * tanhf(drive) +
-                  0.05f * eB[d] * asign;
-        sq += snew[d] * snew[d];
-      }
-      L.racc[tid] += 1.f - 0.1f * sq;
-#pragma unroll
-      for (int d = 0; d < OBS; ++d) L.S[tid][d] = snew[d];
+      for (int e = 0; e < OBS; ++e) drive += eA_d[e] * L.S[cur][env][e];
+      const float snew = 0.97f * L.S[cur][env][dd] +
+                         0.08f * fast_tanh(drive) + 0.05f * eB_d * asign;
+      L.S[cur ^ 1][env][dd] = snew;
+      __syncthreads();  // everyone done reading L.part row before overwrite
+      L.part[dd][env][0] = snew * snew;  // staged for next phase-A reward
     }
     __syncthreads();
   }
 
-  // ---- epilogue: fitness + obs-stat reduction --------------------------
+  // ---- epilogue: last step's reward, fitness, obs-stat reduction -------
   if (tid < ENVS) {
-#pragma unroll
-    for (int d = 0; d < OBS; ++d) {
-      atomicAdd(&L.ostat[d], psum[d]);
-      atomicAdd(&L.ostat[OBS + d], psq[d]);
-    }
+    L.racc[tid] += 1.f - 0.1f * (L.part[0][tid][0] + L.part[1][tid][0] +
+                                 L.part[2][tid][0] + L.part[3][tid][0]);
   }
+  atomicAdd(&L.ostat[dd], psum);
+  atomicAdd(&L.ostat[OBS + dd], psq);
   __syncthreads();
   if (wave == 0) {
     float r = L.racc[lane];
