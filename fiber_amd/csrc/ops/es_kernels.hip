@@ -42,7 +42,9 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // correctly (inf -> tanh=1).
 __device__ inline float fast_tanh(float x) {
   float e = __expf(2.0f * x);
-  return 1.0f - 2.0f / (e + 1.0f);
+  // v_rcp_f32 (1 inst, ~1 ulp) — plain division emits the full IEEE
+  // v_div_scale/v_div_fmas chain (~10 insts), which dominated VALUBusy.
+  return 1.0f - 2.0f * __builtin_amdgcn_rcpf(e + 1.0f);
 }
 
 #define FAM_TAG_NOISE 0x45530001u
@@ -148,13 +150,21 @@ __device__ inline void store_param(PolicyLds* p, int j, float v) {
 // Shared epilogue/prologue state for one member rollout.
 struct RolloutLds {
   PolicyLds pol;
-  alignas(16) __hip_bfloat16 xb[ENVS][S1];  // normalized obs, layer-1 B
+  // xb (layer-1 B operand, live phase A->B) aliases h2 (layer-2 output,
+  // live phase C->D1): their live ranges never overlap, and the union
+  // shrinks the block to ~39 KB => 4 workgroups/CU instead of 3.
+  // (Garbage in xb's K-padding from old h2 values is harmless: the
+  // W1-side padding is zero and old h2 values are finite, never NaN.)
+  union {
+    alignas(16) __hip_bfloat16 xb[ENVS][S1];
+    alignas(16) __hip_bfloat16 h2[ENVS][S2];
+  };
   alignas(16) __hip_bfloat16 h1[ENVS][S2];  // B-operand layer 2
-  alignas(16) __hip_bfloat16 h2[ENVS][S2];  // input to logits
-  float S[2][ENVS][OBS];                    // env state (double-buffered)
-  float part[OBS][ENVS][2];  // logits partials / squared-state partials
-  float racc[ENVS];          // per-env return
-  float ostat[2 * OBS + 1];  // sum, sumsq, count
+  float S[2][ENVS][OBS];      // env state (double-buffered)
+  float lpart[OBS][ENVS][2];  // logits partials (phase D1 -> D2)
+  float sq[OBS][ENVS];        // squared-state partials (D2 -> next A)
+  float racc[ENVS];           // per-env return
+  float ostat[2 * OBS + 1];   // sum, sumsq, count
 };
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -231,9 +241,9 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
       x = fminf(5.f, fmaxf(-5.f, x));
       L.xb[env][dd] = __float2bfloat16(x);
       if (dd == 0 && t > 0) {
-        // reward for step t-1 (squared-state partials staged in L.part)
-        L.racc[env] += 1.f - 0.1f * (L.part[0][env][0] + L.part[1][env][0] +
-                                     L.part[2][env][0] + L.part[3][env][0]);
+        // reward for step t-1 (squared-state partials staged in L.sq)
+        L.racc[env] += 1.f - 0.1f * (L.sq[0][env] + L.sq[1][env] +
+                                     L.sq[2][env] + L.sq[3][env]);
       }
     }
     __syncthreads();
@@ -255,16 +265,18 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
         p0 += __bfloat162float(L.pol.w3[0][h]) * hv;
         p1 += __bfloat162float(L.pol.w3[1][h]) * hv;
       }
-      L.part[dd][env][0] = p0;
-      L.part[dd][env][1] = p1;
+      L.lpart[dd][env][0] = p0;
+      L.lpart[dd][env][1] = p1;
     }
     __syncthreads();
     // ---- phase D2: action + env step (thread (env, d)) ----------------
     {
-      const float l0 = L.pol.b3[0] + L.part[0][env][0] + L.part[1][env][0] +
-                       L.part[2][env][0] + L.part[3][env][0];
-      const float l1 = L.pol.b3[1] + L.part[0][env][1] + L.part[1][env][1] +
-                       L.part[2][env][1] + L.part[3][env][1];
+      const float l0 = L.pol.b3[0] + L.lpart[0][env][0] +
+                       L.lpart[1][env][0] + L.lpart[2][env][0] +
+                       L.lpart[3][env][0];
+      const float l1 = L.pol.b3[1] + L.lpart[0][env][1] +
+                       L.lpart[1][env][1] + L.lpart[2][env][1] +
+                       L.lpart[3][env][1];
       const float asign = (l1 > l0) ? 1.f : -1.f;
       float drive = 0.f;
 #pragma unroll
@@ -272,16 +284,15 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
       const float snew = 0.97f * L.S[cur][env][dd] +
                          0.08f * fast_tanh(drive) + 0.05f * eB_d * asign;
       L.S[cur ^ 1][env][dd] = snew;
-      __syncthreads();  // everyone done reading L.part row before overwrite
-      L.part[dd][env][0] = snew * snew;  // staged for next phase-A reward
+      L.sq[dd][env] = snew * snew;  // staged for next phase-A reward
     }
     __syncthreads();
   }
 
   // ---- epilogue: last step's reward, fitness, obs-stat reduction -------
   if (tid < ENVS) {
-    L.racc[tid] += 1.f - 0.1f * (L.part[0][tid][0] + L.part[1][tid][0] +
-                                 L.part[2][tid][0] + L.part[3][tid][0]);
+    L.racc[tid] += 1.f - 0.1f * (L.sq[0][tid] + L.sq[1][tid] +
+                                 L.sq[2][tid] + L.sq[3][tid]);
   }
   atomicAdd(&L.ostat[dd], psum);
   atomicAdd(&L.ostat[OBS + dd], psq);
