@@ -5,6 +5,7 @@
 // fiber_amd/ops/__init__.py), which keeps the extension a plain hipcc
 // build and the launch path thin.
 
+#include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
 
@@ -20,12 +21,33 @@ extern "C" __global__ void es_rollout_mlp(
     const float*, float, uint32_t, uint32_t, int, int, const float*,
     const float*, const float*, const float*, float*, float*);
 extern "C" __global__ void es_grad(const float*, int, int, int, uint32_t,
-                                   uint32_t, float*);
+                                   uint32_t, int, float*);
 extern "C" __global__ void centered_rank(const float*, int, float*);
 extern "C" __global__ void mlp_policy_forward(const float*, const float*,
                                               int, float*);
 extern "C" __global__ void mfma_gemm64_probe(const float*, const float*,
                                              float*);
+// conv policy pipeline (conv_kernels.hip)
+extern "C" __global__ void es_perturb(const float*, int, int, float,
+                                      uint32_t, uint32_t, int,
+                                      __hip_bfloat16*);
+extern "C" __global__ void conv_env_init(uint32_t, uint32_t, int, float*,
+                                         float*);
+extern "C" __global__ void conv_obsgen(const float*, const float*, uint32_t,
+                                       uint32_t, uint32_t, __hip_bfloat16*);
+extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
+                                       const __hip_bfloat16*, int,
+                                       __hip_bfloat16*);
+extern "C" __global__ void conv_layer2(const __hip_bfloat16*,
+                                       const __hip_bfloat16*, int,
+                                       __hip_bfloat16*);
+extern "C" __global__ void conv_fc(const __hip_bfloat16*,
+                                   const __hip_bfloat16*, int,
+                                   __hip_bfloat16*);
+extern "C" __global__ void conv_head_env(const __hip_bfloat16*,
+                                         const __hip_bfloat16*, int,
+                                         const float*, const float*, float*,
+                                         float*);
 
 static void check(hipError_t err, const char* what) {
   if (err != hipSuccess) {
@@ -50,18 +72,20 @@ static void launch_rollout(uintptr_t theta, double sigma, uint32_t seed,
 }
 
 static void launch_grad(uintptr_t wpair, int pair_begin, int pair_end,
-                        uint32_t seed, uint32_t iter, uintptr_t grad,
-                        uintptr_t stream) {
-  const int jblocks = (NPARAMS_HOST + 3) / 4;
+                        uint32_t seed, uint32_t iter, int nparams,
+                        uintptr_t grad, uintptr_t stream) {
+  const int jblocks = (nparams + 3) / 4;
   const int bx = (jblocks + 255) / 256;
   int pairs = pair_end - pair_begin;
   if (pairs <= 0) return;
-  // Split pairs so enough workgroups exist to fill the chip.
-  int chunks = pairs < 64 ? 1 : 32;
+  // Split pairs so enough workgroups exist to fill the chip (fewer
+  // chunks for large nparams — bx already provides parallelism).
+  int chunks = pairs < 64 ? 1 : (bx >= 32 ? 4 : 32);
   int per_chunk = (pairs + chunks - 1) / chunks;
   hipLaunchKernelGGL(es_grad, dim3(bx, chunks), dim3(256), 0,
                      (hipStream_t)stream, (const float*)wpair, pair_begin,
-                     pair_end, per_chunk, seed, iter, (float*)grad);
+                     pair_end, per_chunk, seed, iter, nparams,
+                     (float*)grad);
   check(hipGetLastError(), "es_grad launch");
 }
 
@@ -90,6 +114,75 @@ static void launch_gemm64_probe(uintptr_t a, uintptr_t b, uintptr_t c,
   check(hipGetLastError(), "mfma_gemm64_probe launch");
 }
 
+// ---- conv pipeline launchers ---------------------------------------------
+
+static void launch_perturb(uintptr_t theta, int nparams, int np_pad,
+                           double sigma, uint32_t seed, uint32_t iter,
+                           int member_offset, int pop, uintptr_t wpert,
+                           uintptr_t stream) {
+  const int bx = 64;  // grid-stride over param blocks
+  hipLaunchKernelGGL(es_perturb, dim3(bx, pop), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)theta, nparams,
+                     np_pad, (float)sigma, seed, iter, member_offset,
+                     (__hip_bfloat16*)wpert);
+  check(hipGetLastError(), "es_perturb launch");
+}
+
+static void launch_conv_env_init(uint32_t seed, uint32_t iter, int nmembers,
+                                 uintptr_t state, uintptr_t racc,
+                                 uintptr_t stream) {
+  const int n = nmembers * 16;
+  hipLaunchKernelGGL(conv_env_init, dim3((n + 255) / 256), dim3(256), 0,
+                     (hipStream_t)stream, seed, iter, nmembers,
+                     (float*)state, (float*)racc);
+  check(hipGetLastError(), "conv_env_init launch");
+}
+
+static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
+                               uint32_t seed, uint32_t iter, uint32_t t,
+                               int nenv_total, uintptr_t obs,
+                               uintptr_t stream) {
+  hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)state,
+                     (const float*)gtab, seed, iter, t,
+                     (__hip_bfloat16*)obs);
+  check(hipGetLastError(), "conv_obsgen launch");
+}
+
+static void launch_conv_forward(uintptr_t wpert, uintptr_t obs,
+                                uintptr_t act1, uintptr_t act2,
+                                uintptr_t act3, int nmembers,
+                                uintptr_t stream) {
+  const int nenv = nmembers * 16;
+  hipLaunchKernelGGL(conv_layer1, dim3(nenv), dim3(256), 0,
+                     (hipStream_t)stream, (const __hip_bfloat16*)wpert,
+                     (const __hip_bfloat16*)obs, nenv,
+                     (__hip_bfloat16*)act1);
+  check(hipGetLastError(), "conv_layer1 launch");
+  hipLaunchKernelGGL(conv_layer2, dim3(nenv), dim3(256), 0,
+                     (hipStream_t)stream, (const __hip_bfloat16*)wpert,
+                     (const __hip_bfloat16*)act1, nenv,
+                     (__hip_bfloat16*)act2);
+  check(hipGetLastError(), "conv_layer2 launch");
+  hipLaunchKernelGGL(conv_fc, dim3(nmembers), dim3(256), 0,
+                     (hipStream_t)stream, (const __hip_bfloat16*)wpert,
+                     (const __hip_bfloat16*)act2, nmembers,
+                     (__hip_bfloat16*)act3);
+  check(hipGetLastError(), "conv_fc launch");
+}
+
+static void launch_conv_head_env(uintptr_t wpert, uintptr_t act3,
+                                 int nmembers, uintptr_t env_A,
+                                 uintptr_t env_B, uintptr_t state,
+                                 uintptr_t racc, uintptr_t stream) {
+  hipLaunchKernelGGL(conv_head_env, dim3(nmembers), dim3(256), 0,
+                     (hipStream_t)stream, (const __hip_bfloat16*)wpert,
+                     (const __hip_bfloat16*)act3, nmembers,
+                     (const float*)env_A, (const float*)env_B,
+                     (float*)state, (float*)racc);
+  check(hipGetLastError(), "conv_head_env launch");
+}
+
 PYBIND11_MODULE(_ops, m) {
   m.doc() = "fiber_amd CDNA4 ES kernels (gfx950)";
   m.attr("NPARAMS") = NPARAMS_HOST;
@@ -102,7 +195,7 @@ PYBIND11_MODULE(_ops, m) {
         py::arg("stream"));
   m.def("es_grad", &launch_grad, py::arg("wpair"), py::arg("pair_begin"),
         py::arg("pair_end"), py::arg("seed"), py::arg("iter"),
-        py::arg("grad"), py::arg("stream"));
+        py::arg("nparams"), py::arg("grad"), py::arg("stream"));
   m.def("centered_rank", &launch_centered_rank, py::arg("f"), py::arg("n"),
         py::arg("out"), py::arg("stream"));
   m.def("mlp_policy_forward", &launch_mlp_forward, py::arg("theta"),
@@ -110,4 +203,25 @@ PYBIND11_MODULE(_ops, m) {
         py::arg("stream"));
   m.def("mfma_gemm64_probe", &launch_gemm64_probe, py::arg("a"),
         py::arg("b"), py::arg("c"), py::arg("stream"));
+
+  m.attr("NP_CONV") = 677686;
+  m.attr("NP_CONV_PAD") = 677688;
+  m.attr("CONV_ENVS") = 16;
+  m.def("es_perturb", &launch_perturb, py::arg("theta"), py::arg("nparams"),
+        py::arg("np_pad"), py::arg("sigma"), py::arg("seed"),
+        py::arg("iter"), py::arg("member_offset"), py::arg("pop"),
+        py::arg("wpert"), py::arg("stream"));
+  m.def("conv_env_init", &launch_conv_env_init, py::arg("seed"),
+        py::arg("iter"), py::arg("nmembers"), py::arg("state"),
+        py::arg("racc"), py::arg("stream"));
+  m.def("conv_obsgen", &launch_conv_obsgen, py::arg("state"),
+        py::arg("gtab"), py::arg("seed"), py::arg("iter"), py::arg("t"),
+        py::arg("nenv_total"), py::arg("obs"), py::arg("stream"));
+  m.def("conv_forward", &launch_conv_forward, py::arg("wpert"),
+        py::arg("obs"), py::arg("act1"), py::arg("act2"), py::arg("act3"),
+        py::arg("nmembers"), py::arg("stream"));
+  m.def("conv_head_env", &launch_conv_head_env, py::arg("wpert"),
+        py::arg("act3"), py::arg("nmembers"), py::arg("env_A"),
+        py::arg("env_B"), py::arg("state"), py::arg("racc"),
+        py::arg("stream"));
 }

@@ -1,0 +1,381 @@
+// CDNA4 kernels for the ConvNet-policy ES path (BASELINE config 4):
+// DQN-shaped policy on 84x84x4 synthetic pixel observations, batched
+// rollouts resident in HBM (288 GB budget), every conv/fc layer on MFMA.
+//
+//   obs 84x84x4 --conv 16@8x8 s4--> 20x20x16 --conv 32@4x4 s2--> 9x9x32
+//       --fc 256--> --head 6-->
+//
+// Data layouts (all channel-last so im2col K-runs are memory-contiguous):
+//   obs  [member][env][y][x][ic=4]    bf16
+//   act1 [member][env][pos=20x20][16] bf16
+//   act2 [member][env][pos=9x9][32]   bf16   (flat k for fc = (y*9+x)*32+oc)
+//   act3 [member][env][256]           bf16
+// Weights: one flat fp32 master theta; es_perturb materializes per-member
+// bf16 perturbed copies (antithetic Philox pairs, same counter scheme as
+// the MLP path) into wpert[member][NP_CONV_PAD].
+//
+// K-ordering of each weight row matches the B-side layout exactly:
+//   W1 row k = (ky*8+kx)*4+ic  -> one K-tile(32) = one kernel row ky
+//   W2 row k = (ky*4+kx)*16+ic -> one K-tile(32) = two kx pixels
+//   W3 row k = flat act2 index
+// so every MFMA B-fragment (8 consecutive k, 16 B) is one aligned
+// contiguous load from obs/act buffers.  E = 16 envs per member makes the
+// fc layer a full 256x16x2592 MFMA GEMM (weights read once per member
+// step, amortized over all 16 envs).
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include "philox.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FAM_TAG_NOISE 0x45530001u
+#define FAM_TAG_ENV 0x45530002u
+#define FAM_TAG_OBS 0x45530003u
+
+__device__ inline float fast_tanh_c(float x) {
+  float e = __expf(2.0f * x);
+  return 1.0f - 2.0f * __builtin_amdgcn_rcpf(e + 1.0f);
+}
+
+// geometry
+#define CIN 4
+#define IMG 84
+#define C1 16
+#define O1 20
+#define C2 32
+#define O2 9
+#define FCU 256
+#define NFLAT (O2 * O2 * C2)  // 2592
+#define NACT 6
+#define CENV 16  // envs per member
+#define SDIM 4
+
+// flat theta offsets (fp32 master / bf16 perturbed share the layout)
+#define COFF_W1 0
+#define COFF_B1 (C1 * 256)                  // 4096
+#define COFF_W2 (COFF_B1 + C1)              // 4112
+#define COFF_B2 (COFF_W2 + C2 * 256)        // 12304
+#define COFF_W3 (COFF_B2 + C2)              // 12336
+#define COFF_B3 (COFF_W3 + FCU * NFLAT)     // 675888
+#define COFF_W4 (COFF_B3 + FCU)             // 676144
+#define COFF_B4 (COFF_W4 + NACT * FCU)      // 677680
+#define NP_CONV (COFF_B4 + NACT)            // 677686
+#define NP_CONV_PAD 677688                  // 16B-aligned member stride
+
+// ---------------------------------------------------------------------------
+// es_perturb: wpert[member][NP_CONV_PAD] = bf16(theta +/- sigma*eps_pair)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void es_perturb(const float* __restrict__ theta,
+                                      int nparams, int np_pad, float sigma,
+                                      uint32_t seed, uint32_t iter,
+                                      int member_offset,
+                                      __hip_bfloat16* __restrict__ wpert) {
+  const int member = member_offset + blockIdx.y;
+  const uint32_t pair = (uint32_t)(member >> 1);
+  const float sgn = (member & 1) ? -sigma : sigma;
+  __hip_bfloat16* out = wpert + (size_t)blockIdx.y * np_pad;
+  const int jb0 = blockIdx.x * blockDim.x + threadIdx.x;
+  const int stride = gridDim.x * blockDim.x;
+  for (int jb = jb0; jb * 4 < nparams; jb += stride) {
+    float z[4];
+    fam_normal4(seed, iter, pair, (uint32_t)jb, FAM_TAG_NOISE, 0u, z);
+    const int j0 = jb * 4;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int j = j0 + u;
+      if (j < nparams) out[j] = __float2bfloat16(theta[j] + sgn * z[u]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_env_init: state[b][e][d] = 0.3*z  (member-independent, like MLP)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void conv_env_init(uint32_t seed, uint32_t iter,
+                                         int nmembers,
+                                         float* __restrict__ state,
+                                         float* __restrict__ racc) {
+  const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= nmembers * CENV) return;
+  const int e = idx % CENV;
+  float z[4];
+  fam_normal4(seed, iter, (uint32_t)e, 0u, FAM_TAG_ENV, 0u, z);
+#pragma unroll
+  for (int d = 0; d < SDIM; ++d) state[idx * SDIM + d] = 0.3f * z[d];
+  racc[idx] = 0.f;
+}
+
+// ---------------------------------------------------------------------------
+// conv_obsgen: obs[b][e][y][x][ic] = 0.3*noise(e,t,pos) +
+//                                    state[b][e][ic] * gtab[y][x]
+// one workgroup per (member, env); channel-last writes, 4 bf16 per pos.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
+                                       const float* __restrict__ gtab,
+                                       uint32_t seed, uint32_t iter,
+                                       uint32_t t,
+                                       __hip_bfloat16* __restrict__ obs) {
+  const int be = blockIdx.x;  // member*CENV + env
+  const int e = be % CENV;
+  __hip_bfloat16* out = obs + (size_t)be * (IMG * IMG * CIN);
+  float s[SDIM];
+#pragma unroll
+  for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
+  // each position needs CIN=4 values; one philox draw per position
+  for (int p = threadIdx.x; p < IMG * IMG; p += blockDim.x) {
+    float z[4];
+    fam_normal4(seed, iter, (uint32_t)e, (uint32_t)p, FAM_TAG_OBS, t, z);
+    const float g = gtab[p];
+    union {
+      __hip_bfloat16 h[4];
+      unsigned long long u;
+    } pk;
+#pragma unroll
+    for (int c = 0; c < CIN; ++c)
+      pk.h[c] = __float2bfloat16(0.3f * z[c] + s[c] * g);
+    *reinterpret_cast<unsigned long long*>(&out[p * CIN]) = pk.u;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_layer1: act1 = tanh(conv(obs, W1) + b1); one wg per (member, env).
+// M=16 (one tile), N=400 (25 tiles), K=256 (8 tiles = 8 kernel rows).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+conv_layer1(const __hip_bfloat16* __restrict__ wpert,
+            const __hip_bfloat16* __restrict__ obs, int nenv_total,
+            __hip_bfloat16* __restrict__ act1) {
+  __shared__ alignas(16) __hip_bfloat16 w1[C1][256];
+  __shared__ float b1[C1];
+  const int be = blockIdx.x;
+  const int member = be / CENV;
+  const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
+  const __hip_bfloat16* ob = obs + (size_t)be * (IMG * IMG * CIN);
+  __hip_bfloat16* out = act1 + (size_t)be * (O1 * O1 * C1);
+
+  const int tid = threadIdx.x;
+  for (int i = tid; i < C1 * 256 / 8; i += blockDim.x) {
+    reinterpret_cast<bf16x8*>(&w1[0][0])[i] =
+        reinterpret_cast<const bf16x8*>(wm + COFF_W1)[i];
+  }
+  if (tid < C1) b1[tid] = __bfloat162float(wm[COFF_B1 + tid]);
+  __syncthreads();
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kgrp = lane >> 4;
+  const int arow = lane & 15;  // output channel
+
+  bf16x8 afrag[8];
+#pragma unroll
+  for (int kk = 0; kk < 8; ++kk)
+    afrag[kk] =
+        *reinterpret_cast<const bf16x8*>(&w1[arow][kk * 32 + kgrp * 8]);
+
+  for (int nt = wave; nt < 25; nt += 4) {
+    const int pos = nt * 16 + (lane & 15);
+    const int oy = pos / O1, ox = pos % O1;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) {  // ky = kk
+      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &ob[(((oy * 4 + kk) * IMG) + ox * 4 + kgrp * 2) * CIN]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bfrag, acc,
+                                                    0, 0, 0);
+    }
+    const int drow = kgrp * 4;  // output-channel base for this lane's D
+    union {
+      __hip_bfloat16 h[4];
+      unsigned long long u;
+    } pk;
+#pragma unroll
+    for (int ri = 0; ri < 4; ++ri)
+      pk.h[ri] = __float2bfloat16(fast_tanh_c(acc[ri] + b1[drow + ri]));
+    *reinterpret_cast<unsigned long long*>(&out[pos * C1 + drow]) = pk.u;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_layer2: act2 = tanh(conv(act1, W2) + b2); one wg per (member, env).
+// M=32 (2 tiles), N=81 (6 tiles, padded), K=256 (8 tiles = 2 pixels each).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+conv_layer2(const __hip_bfloat16* __restrict__ wpert,
+            const __hip_bfloat16* __restrict__ act1, int nenv_total,
+            __hip_bfloat16* __restrict__ act2) {
+  __shared__ alignas(16) __hip_bfloat16 w2[C2][256];
+  __shared__ float b2[C2];
+  const int be = blockIdx.x;
+  const int member = be / CENV;
+  const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
+  const __hip_bfloat16* in = act1 + (size_t)be * (O1 * O1 * C1);
+  __hip_bfloat16* out = act2 + (size_t)be * NFLAT;
+
+  const int tid = threadIdx.x;
+  for (int i = tid; i < C2 * 256 / 8; i += blockDim.x) {
+    reinterpret_cast<bf16x8*>(&w2[0][0])[i] =
+        reinterpret_cast<const bf16x8*>(wm + COFF_W2)[i];
+  }
+  if (tid < C2) b2[tid] = __bfloat162float(wm[COFF_B2 + tid]);
+  __syncthreads();
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kgrp = lane >> 4;
+
+  // 12 tile jobs: job = mt*6 + ntile
+  for (int job = wave; job < 12; job += 4) {
+    const int mt = job / 6, nt = job % 6;
+    const int arow = mt * 16 + (lane & 15);
+    int pos = nt * 16 + (lane & 15);
+    const bool valid = pos < O2 * O2;
+    if (!valid) pos = 0;
+    const int oy = pos / O2, ox = pos % O2;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) {
+      // K-tile kk covers pixels (ky = kk/2, kx = (kk%2)*2 + {0,1})
+      const int ky = kk >> 1;
+      const int kx0 = (kk & 1) * 2;
+      const int pix = ((oy * 2 + ky) * O1) + ox * 2 + kx0 + (kgrp >> 1);
+      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          &w2[arow & 31][kk * 32 + kgrp * 8]);
+      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &in[pix * C1 + (kgrp & 1) * 8]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0,
+                                                    0, 0);
+    }
+    const int drow = mt * 16 + kgrp * 4;
+    const int dpos = nt * 16 + (lane & 15);
+    if (dpos < O2 * O2) {
+      union {
+        __hip_bfloat16 h[4];
+        unsigned long long u;
+      } pk;
+#pragma unroll
+      for (int ri = 0; ri < 4; ++ri)
+        pk.h[ri] = __float2bfloat16(
+            fast_tanh_c(acc[ri] + b2[(drow + ri) & 31]));
+      *reinterpret_cast<unsigned long long*>(
+          &out[dpos * C2 + (drow & 31)]) = pk.u;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_fc: act3 = tanh(W3 @ act2_flat + b3); one wg per MEMBER.
+// M=256 (16 tiles), N=16 (the member's envs), K=2592 (81 tiles).
+// Weights are read straight from HBM (single use), activations too.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+conv_fc(const __hip_bfloat16* __restrict__ wpert,
+        const __hip_bfloat16* __restrict__ act2, int nmembers,
+        __hip_bfloat16* __restrict__ act3) {
+  const int member = blockIdx.x;
+  const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
+  const __hip_bfloat16* w3 = wm + COFF_W3;
+  const __hip_bfloat16* in = act2 + (size_t)member * CENV * NFLAT;
+  __hip_bfloat16* out = act3 + (size_t)member * CENV * FCU;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kgrp = lane >> 4;
+  const int env = lane & 15;  // B column
+
+  for (int mt = wave; mt < 16; mt += 4) {
+    const int arow = mt * 16 + (lane & 15);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int kk = 0; kk < 81; ++kk) {
+      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          &w3[(size_t)arow * NFLAT + kk * 32 + kgrp * 8]);
+      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &in[(size_t)env * NFLAT + kk * 32 + kgrp * 8]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0,
+                                                    0, 0);
+    }
+    const int drow = mt * 16 + kgrp * 4;
+    union {
+      __hip_bfloat16 h[4];
+      unsigned long long u;
+    } pk;
+#pragma unroll
+    for (int ri = 0; ri < 4; ++ri) {
+      const float bias = __bfloat162float(wm[COFF_B3 + drow + ri]);
+      pk.h[ri] = __float2bfloat16(fast_tanh_c(acc[ri] + bias));
+    }
+    // D col = env, rows drow..drow+3 -> act3[env][drow..]
+    *reinterpret_cast<unsigned long long*>(&out[env * FCU + drow]) = pk.u;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_head_env: logits -> argmax action -> synthetic env step -> reward.
+// One wg per member; one wave per 4 envs (16 lanes per env).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void conv_head_env(
+    const __hip_bfloat16* __restrict__ wpert,
+    const __hip_bfloat16* __restrict__ act3, int nmembers,
+    const float* __restrict__ env_A, const float* __restrict__ env_B,
+    float* __restrict__ state, float* __restrict__ racc) {
+  const int member = blockIdx.x;
+  const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
+  const __hip_bfloat16* a3 = act3 + (size_t)member * CENV * FCU;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int env = wave * 4 + (lane >> 4);       // 0..15
+  const int usub = lane & 15;                   // 16 unit groups of 16
+  const int be = member * CENV + env;
+
+  float part[NACT];
+#pragma unroll
+  for (int a = 0; a < NACT; ++a) part[a] = 0.f;
+  for (int uu = 0; uu < 16; ++uu) {
+    const int u = usub * 16 + uu;
+    const float hv = __bfloat162float(a3[env * FCU + u]);
+#pragma unroll
+    for (int a = 0; a < NACT; ++a)
+      part[a] += __bfloat162float(wm[COFF_W4 + a * FCU + u]) * hv;
+  }
+  // reduce across the 16 lanes of this env group
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) {
+#pragma unroll
+    for (int a = 0; a < NACT; ++a)
+      part[a] += __shfl_down(part[a], off, 16);
+  }
+  if (usub == 0) {
+    int best = 0;
+    float bestv = part[0] + __bfloat162float(wm[COFF_B4]);
+#pragma unroll
+    for (int a = 1; a < NACT; ++a) {
+      const float v = part[a] + __bfloat162float(wm[COFF_B4 + a]);
+      if (v > bestv) {
+        bestv = v;
+        best = a;
+      }
+    }
+    const float force = ((float)best - 2.5f) * 0.4f;  // in [-1, 1]
+    float s[SDIM], snew[SDIM];
+#pragma unroll
+    for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
+    float sq = 0.f;
+#pragma unroll
+    for (int d = 0; d < SDIM; ++d) {
+      float drive = 0.f;
+#pragma unroll
+      for (int e2 = 0; e2 < SDIM; ++e2) drive += env_A[d * SDIM + e2] * s[e2];
+      snew[d] = 0.97f * s[d] + 0.08f * fast_tanh_c(drive) +
+                0.05f * env_B[d] * force;
+      sq += snew[d] * snew[d];
+    }
+#pragma unroll
+    for (int d = 0; d < SDIM; ++d) state[be * SDIM + d] = snew[d];
+    racc[be] += 1.f - 0.1f * sq;
+  }
+}

@@ -315,10 +315,10 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
 extern "C" __global__ void es_grad(const float* __restrict__ wpair,
                                    int pair_begin, int pair_end,
                                    int pairs_per_chunk, uint32_t seed,
-                                   uint32_t iter,
+                                   uint32_t iter, int nparams,
                                    float* __restrict__ grad) {
   const int jb = blockIdx.x * blockDim.x + threadIdx.x;
-  if (jb * 4 >= NPARAMS) return;
+  if (jb * 4 >= nparams) return;
   const int chunk_begin = pair_begin + blockIdx.y * pairs_per_chunk;
   const int chunk_end = min(chunk_begin + pairs_per_chunk, pair_end);
   float acc[4] = {0.f, 0.f, 0.f, 0.f};
@@ -334,7 +334,7 @@ extern "C" __global__ void es_grad(const float* __restrict__ wpair,
 #pragma unroll
   for (int u = 0; u < 4; ++u) {
     const int j = jb * 4 + u;
-    if (j < NPARAMS && (acc[u] != 0.f)) atomicAdd(&grad[j], acc[u]);
+    if (j < nparams && (acc[u] != 0.f)) atomicAdd(&grad[j], acc[u]);
   }
 }
 

@@ -1,0 +1,269 @@
+"""ConvNet-policy ES engine (BASELINE config 4).
+
+DQN-shaped policy on 84x84x4 synthetic pixel observations; population of
+perturbed weight sets materialized in HBM (bf16, ~1.4 MB/member — the
+288 GB budget makes thousand-member populations cheap); each rollout
+step is a 4-kernel MFMA pipeline (obsgen -> conv1 -> conv2 -> fc ->
+head+env) over pop x 16 envs.  Shares the Philox noise scheme, the
+centered-rank/gradient kernels and the Ring collectives with the MLP
+engine.
+"""
+
+import dataclasses
+
+import torch
+
+from .. import ops
+from .engine import make_env_params
+
+
+def conv_dims():
+    o = ops._require_ops()
+    return o.NP_CONV, o.NP_CONV_PAD, o.CONV_ENVS
+
+
+# flat layout offsets (mirror conv_kernels.hip)
+OFF = {
+    "w1": (0, (16, 8, 8, 4)),       # stored [oc][(ky*8+kx)*4+ic]
+    "b1": (4096, (16,)),
+    "w2": (4112, (32, 4, 4, 16)),   # [oc][(ky*4+kx)*16+ic]
+    "b2": (12304, (32,)),
+    "w3": (12336, (256, 2592)),     # [out][flat act2 (y*9+x)*32+oc]
+    "b3": (675888, (256,)),
+    "w4": (676144, (6, 256)),
+    "b4": (677680, (6,)),
+}
+NP_CONV = 677686
+
+
+@dataclasses.dataclass
+class ConvESConfig:
+    pop_per_gpu: int = 512        # must be even
+    horizon: int = 64
+    sigma: float = 0.02
+    lr: float = 0.01
+    seed: int = 4321
+    adam_beta1: float = 0.9
+    adam_beta2: float = 0.999
+    adam_eps: float = 1e-8
+    envs_per_member: int = 16
+
+
+def init_conv_theta(seed, device):
+    gen = torch.Generator().manual_seed(seed)
+    theta = torch.zeros(NP_CONV)
+
+    def put(name, scale):
+        off, shape = OFF[name]
+        n = 1
+        for s in shape:
+            n *= s
+        w = torch.randn(n, generator=gen) * scale
+        theta[off : off + n] = w
+
+    put("w1", (8 * 8 * 4) ** -0.5)
+    put("w2", (4 * 4 * 16) ** -0.5)
+    put("w3", 2592 ** -0.5)
+    put("w4", 256 ** -0.5)
+    return theta.to(device).contiguous()
+
+
+def make_gtab(seed, device):
+    """Fixed spatial modulation pattern g[y][x] for the synthetic obs."""
+    gen = torch.Generator().manual_seed(seed ^ 0x6A7)
+    y, x = torch.meshgrid(
+        torch.linspace(-1, 1, 84), torch.linspace(-1, 1, 84),
+        indexing="ij",
+    )
+    phase = torch.rand(2, generator=gen) * 6.28
+    g = torch.sin(3.0 * x + phase[0]) * torch.cos(2.0 * y + phase[1])
+    return g.reshape(-1).to(device).contiguous()
+
+
+class ConvESEngine:
+    def __init__(self, config: ConvESConfig, ctx=None, device=None):
+        o = ops._require_ops()
+        self.cfg = config
+        self.ctx = ctx
+        self.rank = ctx.rank if ctx else 0
+        self.world = ctx.size if ctx else 1
+        if device is None:
+            device = torch.device("cuda", 0)
+        self.device = device
+        if config.pop_per_gpu % 2:
+            raise ValueError("pop_per_gpu must be even")
+        self.pop_total = config.pop_per_gpu * self.world
+        E = config.envs_per_member
+        assert E == o.CONV_ENVS, "envs_per_member fixed by the kernel"
+        pop = config.pop_per_gpu
+
+        self.theta = init_conv_theta(config.seed, device)
+        self.adam_m = torch.zeros_like(self.theta)
+        self.adam_v = torch.zeros_like(self.theta)
+        self.t_step = 0
+        self.env_A, self.env_B = make_env_params(config.seed, device)
+        self.gtab = make_gtab(config.seed, device)
+
+        bf = torch.bfloat16
+        self.wpert = torch.empty(pop, o.NP_CONV_PAD, dtype=bf, device=device)
+        self.obs = torch.empty(pop * E, 84 * 84 * 4, dtype=bf, device=device)
+        self.act1 = torch.empty(pop * E, 20 * 20 * 16, dtype=bf,
+                                device=device)
+        self.act2 = torch.empty(pop * E, 2592, dtype=bf, device=device)
+        self.act3 = torch.empty(pop * E, 256, dtype=bf, device=device)
+        self.state = torch.empty(pop * E, 4, dtype=torch.float32,
+                                 device=device)
+        self.racc = torch.empty(pop, E, dtype=torch.float32, device=device)
+        self._fitness_all = torch.empty(self.pop_total, device=device)
+
+    def _stream(self):
+        return torch.cuda.current_stream().cuda_stream
+
+    def rollout(self, iteration):
+        """Run the shard's rollouts; returns fitness[pop] on device."""
+        o = ops._require_ops()
+        cfg = self.cfg
+        pop = cfg.pop_per_gpu
+        member_offset = self.rank * pop
+        stream = self._stream()
+        o.es_perturb(self.theta.data_ptr(), NP_CONV, o.NP_CONV_PAD,
+                     cfg.sigma, cfg.seed, iteration, member_offset, pop,
+                     self.wpert.data_ptr(), stream)
+        o.conv_env_init(cfg.seed, iteration, pop, self.state.data_ptr(),
+                        self.racc.data_ptr(), stream)
+        for t in range(cfg.horizon):
+            o.conv_obsgen(self.state.data_ptr(), self.gtab.data_ptr(),
+                          cfg.seed, iteration, t, pop * cfg.envs_per_member,
+                          self.obs.data_ptr(), stream)
+            o.conv_forward(self.wpert.data_ptr(), self.obs.data_ptr(),
+                           self.act1.data_ptr(), self.act2.data_ptr(),
+                           self.act3.data_ptr(), pop, stream)
+            o.conv_head_env(self.wpert.data_ptr(), self.act3.data_ptr(),
+                            pop, self.env_A.data_ptr(),
+                            self.env_B.data_ptr(), self.state.data_ptr(),
+                            self.racc.data_ptr(), stream)
+        return self.racc.mean(dim=1)
+
+    def step(self, iteration=None):
+        cfg = self.cfg
+        if iteration is None:
+            iteration = self.t_step
+        pop = cfg.pop_per_gpu
+        member_offset = self.rank * pop
+
+        fitness = self.rollout(iteration).contiguous()
+        if self.ctx is not None:
+            self.ctx.all_gather_into(self._fitness_all, fitness)
+            fitness_all = self._fitness_all
+        else:
+            fitness_all = fitness
+
+        ranks = ops.centered_rank(fitness_all)
+        wpair = (ranks[0::2] - ranks[1::2]).contiguous()
+        grad = ops.es_grad(wpair, member_offset // 2,
+                           (member_offset + pop) // 2, cfg.seed, iteration,
+                           self.device, nparams=NP_CONV)
+        if self.ctx is not None:
+            self.ctx.allreduce(grad)
+        grad /= float(self.pop_total) * cfg.sigma
+
+        self.t_step += 1
+        b1, b2 = cfg.adam_beta1, cfg.adam_beta2
+        self.adam_m.mul_(b1).add_(grad, alpha=1 - b1)
+        self.adam_v.mul_(b2).addcmul_(grad, grad, value=1 - b2)
+        mhat = self.adam_m / (1 - b1 ** self.t_step)
+        vhat = self.adam_v / (1 - b2 ** self.t_step)
+        self.theta.add_(cfg.lr * mhat / (vhat.sqrt() + cfg.adam_eps))
+
+        return {
+            "fitness_mean": float(fitness_all.mean()),
+            "fitness_max": float(fitness_all.max()),
+            "grad_norm": float(grad.norm()),
+            "rollouts": self.pop_total * cfg.envs_per_member,
+            "env_steps": self.pop_total * cfg.envs_per_member * cfg.horizon,
+        }
+
+
+# ---------------------------------------------------------------------------
+# fp32 torch reference (bf16 rounding at the kernel's rounding points) —
+# numerics oracle for tests/gpu/test_conv.py.
+# ---------------------------------------------------------------------------
+
+
+def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
+                           env_A, env_B, gtab):
+    import numpy as np
+
+    from . import philox_ref
+
+    def bf(t):
+        return t.to(torch.bfloat16).to(torch.float32)
+
+    theta = theta.detach().cpu()
+    env_A = env_A.detach().cpu()
+    env_B = env_B.detach().cpu()
+    gtab = gtab.detach().cpu().reshape(84, 84)
+    E = 16
+    fitness = []
+
+    # member-independent per-env noise streams
+    def obs_noise(t):
+        # [E][84*84][4]
+        outs = []
+        for e in range(E):
+            p = np.arange(84 * 84, dtype=np.uint32)
+            z = philox_ref.normal4(seed, iteration,
+                                   np.uint32(e) * np.ones_like(p), p,
+                                   np.uint32(0x45530003), np.uint32(t))
+            outs.append(torch.from_numpy(z))
+        return torch.stack(outs)
+
+    s0 = torch.from_numpy(
+        philox_ref.env_init_state(seed, iteration, E)
+    )
+
+    for m in members:
+        pair = m // 2
+        sgn = -sigma if (m % 2) else sigma
+        eps = torch.from_numpy(
+            philox_ref.noise_for_pair(seed, iteration, pair, NP_CONV)
+        )
+        th = bf(theta + sgn * eps)  # wpert materializes in bf16
+
+        def get(name):
+            off, shape = OFF[name]
+            n = 1
+            for s in shape:
+                n *= s
+            return th[off : off + n].view(*shape)
+
+        # weight k-orders -> torch conv layout [oc][ic][ky][kx]
+        w1 = get("w1").permute(0, 3, 1, 2).contiguous()
+        b1 = get("b1")
+        w2 = get("w2").permute(0, 3, 1, 2).contiguous()
+        b2 = get("b2")
+        w3, b3 = get("w3"), get("b3")
+        w4, b4 = get("w4"), get("b4")
+
+        s = s0.clone()
+        racc = torch.zeros(E)
+        for t in range(horizon):
+            z = obs_noise(t)  # [E][7056][4]
+            obs = 0.3 * z + s[:, None, :] * gtab.reshape(1, -1, 1)
+            obs = bf(obs)
+            x = obs.reshape(E, 84, 84, 4).permute(0, 3, 1, 2)
+            h1 = bf(torch.tanh(
+                torch.nn.functional.conv2d(x, w1, b1, stride=4)))
+            h2 = bf(torch.tanh(
+                torch.nn.functional.conv2d(h1, w2, b2, stride=2)))
+            flat = h2.permute(0, 2, 3, 1).reshape(E, -1)  # (y*9+x)*32+oc
+            h3 = bf(torch.tanh(flat @ w3.T + b3))
+            logits = h3 @ w4.T + b4
+            action = logits.argmax(dim=1).float()
+            force = (action - 2.5) * 0.4
+            drive = torch.tanh(s @ env_A.T)
+            snew = 0.97 * s + 0.08 * drive + 0.05 * env_B * force[:, None]
+            racc += 1.0 - 0.1 * (snew * snew).sum(dim=1)
+            s = snew
+        fitness.append(float(racc.mean()))
+    return torch.tensor(fitness)

@@ -70,14 +70,17 @@ def es_rollout_mlp(theta, sigma, seed, iteration, horizon, member_offset,
     return fitness, obs_stat
 
 
-def es_grad(wpair, pair_begin, pair_end, seed, iteration, device):
+def es_grad(wpair, pair_begin, pair_end, seed, iteration, device,
+            nparams=None):
     """Noise-weighted gradient over local pairs; eps regenerated on-chip."""
     ops = _require_ops()
     _check(wpair, "wpair")
-    grad = torch.zeros(NPARAMS, dtype=torch.float32, device=device)
+    if nparams is None:
+        nparams = NPARAMS
+    grad = torch.zeros(nparams, dtype=torch.float32, device=device)
     ops.es_grad(wpair.data_ptr(), int(pair_begin), int(pair_end),
                 int(seed) & 0xFFFFFFFF, int(iteration) & 0xFFFFFFFF,
-                grad.data_ptr(), _stream())
+                int(nparams), grad.data_ptr(), _stream())
     return grad
 
 

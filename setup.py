@@ -39,6 +39,8 @@ class build_ext(_build_ext):
         out = os.path.join(ROOT, "fiber_amd", "_ops" + ext_suffix())
         srcs = [
             os.path.join(ROOT, "fiber_amd", "csrc", "ops", "es_kernels.hip"),
+            os.path.join(ROOT, "fiber_amd", "csrc", "ops",
+                         "conv_kernels.hip"),
             os.path.join(ROOT, "fiber_amd", "csrc", "ops", "bindings.cpp"),
         ]
         hdrs = [
