@@ -98,11 +98,16 @@ class Server:
         self._lock = threading.Lock()
         self._shutdown = threading.Event()
 
+    def bind(self):
+        self._sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        self._sock.bind(self.address)
+        self._sock.listen(128)
+        self._sock.settimeout(0.2)
+
     def serve_forever(self):
-        sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
-        sock.bind(self.address)
-        sock.listen(128)
-        sock.settimeout(0.2)
+        if getattr(self, "_sock", None) is None:
+            self.bind()
+        sock = self._sock
         while not self._shutdown.is_set():
             try:
                 conn, _ = sock.accept()
@@ -194,6 +199,7 @@ _EXPOSED_DUNDERS = {
 
 def _run_server(address, registry, report_conn):
     server = Server(address, registry)
+    server.bind()  # bind BEFORE advertising so clients can always connect
     report_conn.send(address)
     server.serve_forever()
 
