@@ -28,9 +28,16 @@ def main():
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=20)
     parser.add_argument("--warmup", type=int, default=3)
-    parser.add_argument("--pop-per-gpu", type=int, default=4096)
-    parser.add_argument("--horizon", type=int, default=256)
+    parser.add_argument("--pop-per-gpu", type=int, default=None)
+    parser.add_argument("--horizon", type=int, default=None)
+    parser.add_argument("--model", choices=["mlp", "conv"], default="mlp",
+                        help="flagship MLP config (default) or the "
+                             "ConvNet pixel-policy config")
     args = parser.parse_args()
+    if args.pop_per_gpu is None:
+        args.pop_per_gpu = 4096 if args.model == "mlp" else 512
+    if args.horizon is None:
+        args.horizon = 256 if args.model == "mlp" else 64
 
     import torch
 
@@ -48,10 +55,19 @@ def main():
         ctx = RingContext(rank, world, backend="nccl", device=device)
         ctx.init()
 
-    from fiber_amd.es import ESConfig, ESEngine
+    if args.model == "mlp":
+        from fiber_amd.es import ESConfig, ESEngine
 
-    cfg = ESConfig(pop_per_gpu=args.pop_per_gpu, horizon=args.horizon)
-    engine = ESEngine(cfg, ctx=ctx, device=device)
+        cfg = ESConfig(pop_per_gpu=args.pop_per_gpu, horizon=args.horizon)
+        engine = ESEngine(cfg, ctx=ctx, device=device)
+        model_name = "es-mlp-obs4-h64x64-act2"
+    else:
+        from fiber_amd.es.conv_policy import ConvESConfig, ConvESEngine
+
+        cfg = ConvESConfig(pop_per_gpu=args.pop_per_gpu,
+                           horizon=args.horizon)
+        engine = ConvESEngine(cfg, ctx=ctx, device=device)
+        model_name = "es-conv-84x84x4-dqn-act6"
 
     for i in range(args.warmup):
         engine.step(iteration=i)
@@ -93,7 +109,7 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "es-mlp-obs4-h64x64-act2",
+                "model": model_name,
                 "global_batch": cfg.pop_per_gpu * world
                 * cfg.envs_per_member,
                 "seq_len": args.horizon,

@@ -40,9 +40,16 @@ class TestConvPipeline:
                                           conv_policy.NP_CONV)
             )
             sgn = -0.02 if m % 2 else 0.02
-            want = (theta.cpu() + sgn * eps).to(torch.bfloat16)
-            got = wpert[m, : conv_policy.NP_CONV].cpu()
-            assert torch.equal(got, want)
+            want = (theta.cpu() + sgn * eps).to(torch.bfloat16).float()
+            got = wpert[m, : conv_policy.NP_CONV].cpu().float()
+            # device libm (sinf/cosf, fma contraction) differs from numpy
+            # by ~1 ulp fp32, which can flip the bf16 rounding of a few
+            # of the 677k values — require near-total agreement instead
+            # of bitwise equality.
+            close = torch.isclose(got, want, atol=1e-3, rtol=2e-2)
+            frac = close.float().mean().item()
+            assert frac > 0.9999, frac
+            assert (got - want).abs().max().item() < 5e-3
 
     def test_rollout_vs_fp32_reference(self):
         device = torch.device("cuda")
