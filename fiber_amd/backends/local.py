@@ -65,12 +65,34 @@ class Backend(BackendABC):
             env["CUDA_VISIBLE_DEVICES"] = dev_str
         # Keep dmabuf IPC mode for cross-process HIP tensor sharing.
         env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-        proc = subprocess.Popen(job_spec.command, env=env)
-        with self._lock:
-            self._job_seq += 1
-            jid = "local-%d-%d" % (proc.pid, self._job_seq)
+        conf = fam_config.get_object()
+        log_path = None
+        stdout = stderr = None
+        if not conf.merge_output:
+            # Capture job output so get_job_logs works (reference parity:
+            # docker/k8s backends expose container/pod logs).
+            with self._lock:
+                self._job_seq += 1
+                seq = self._job_seq
+            log_path = os.path.join(
+                conf.ipc_dir,
+                "fam-job-%d-%d.log" % (os.getpid(), seq),
+            )
+            stdout = open(log_path, "ab")
+            stderr = subprocess.STDOUT
+        else:
+            with self._lock:
+                self._job_seq += 1
+                seq = self._job_seq
+        proc = subprocess.Popen(
+            job_spec.command, env=env, stdout=stdout, stderr=stderr
+        )
+        if stdout is not None:
+            stdout.close()
+        jid = "local-%d-%d" % (proc.pid, seq)
         job = Job(proc, jid)
         job.devices = devices
+        job.log_path = log_path
         return job
 
     def get_job_status(self, job):
@@ -84,7 +106,19 @@ class Backend(BackendABC):
         return job.data.poll()
 
     def get_job_logs(self, job):
+        path = getattr(job, "log_path", None)
+        if path and os.path.exists(path):
+            with open(path, "rb") as fh:
+                return fh.read().decode(errors="replace")
         return ""
+
+    def cleanup_job(self, job):
+        path = getattr(job, "log_path", None)
+        if path:
+            try:
+                os.unlink(path)
+            except OSError:
+                pass
 
     def wait_for_job(self, job, timeout):
         try:

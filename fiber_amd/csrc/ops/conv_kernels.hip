@@ -127,7 +127,7 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
   // each position needs CIN=4 values; one philox draw per position
   for (int p = threadIdx.x; p < IMG * IMG; p += blockDim.x) {
     float z[4];
-    fam_normal4(seed, iter, (uint32_t)e, (uint32_t)p, FAM_TAG_OBS, t, z);
+    fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)p, FAM_TAG_OBS, t, z);
     const float g = gtab[p];
     union {
       __hip_bfloat16 h[4];
@@ -135,7 +135,7 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
     } pk;
 #pragma unroll
     for (int c = 0; c < CIN; ++c)
-      pk.h[c] = __float2bfloat16(0.3f * z[c] + s[c] * g);
+      pk.h[c] = __float2bfloat16(0.52f * z[c] + s[c] * g);
     *reinterpret_cast<unsigned long long*>(&out[p * CIN]) = pk.u;
   }
 }

@@ -53,6 +53,21 @@ __device__ __host__ inline fam_uint4 philox4x32_10(uint32_t k0, uint32_t k1,
   return {c0, c1, c2, c3};
 }
 
+// 4 uniforms in [-1, 1) from one philox draw (no transcendentals —
+// used for the pixel-observation noise where gaussianity is not needed
+// and Box-Muller's log/sqrt/sincos dominated the obsgen kernel).
+__device__ __host__ inline void fam_uniform4(uint32_t k0, uint32_t k1,
+                                             uint32_t c0, uint32_t c1,
+                                             uint32_t c2, uint32_t c3,
+                                             float z[4]) {
+  fam_uint4 u = philox4x32_10(k0, k1, c0, c1, c2, c3);
+  const float inv31 = 4.656612873077393e-10f;  // 2^-31
+  z[0] = (float)(int32_t)u.x * inv31;
+  z[1] = (float)(int32_t)u.y * inv31;
+  z[2] = (float)(int32_t)u.z * inv31;
+  z[3] = (float)(int32_t)u.w * inv31;
+}
+
 // 4 standard normals from one philox draw (two Box-Muller pairs).
 // u in (0,1]: (x + 1) * 2^-32.
 __device__ __host__ inline void fam_normal4(uint32_t k0, uint32_t k1,

@@ -206,15 +206,15 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
     E = 16
     fitness = []
 
-    # member-independent per-env noise streams
+    # member-independent per-env noise streams (uniform, see fam_uniform4)
     def obs_noise(t):
         # [E][84*84][4]
         outs = []
         for e in range(E):
             p = np.arange(84 * 84, dtype=np.uint32)
-            z = philox_ref.normal4(seed, iteration,
-                                   np.uint32(e) * np.ones_like(p), p,
-                                   np.uint32(0x45530003), np.uint32(t))
+            z = philox_ref.uniform4(seed, iteration,
+                                    np.uint32(e) * np.ones_like(p), p,
+                                    np.uint32(0x45530003), np.uint32(t))
             outs.append(torch.from_numpy(z))
         return torch.stack(outs)
 
@@ -249,7 +249,7 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
         racc = torch.zeros(E)
         for t in range(horizon):
             z = obs_noise(t)  # [E][7056][4]
-            obs = 0.3 * z + s[:, None, :] * gtab.reshape(1, -1, 1)
+            obs = 0.52 * z + s[:, None, :] * gtab.reshape(1, -1, 1)
             obs = bf(obs)
             x = obs.reshape(E, 84, 84, 4).permute(0, 3, 1, 2)
             h1 = bf(torch.tanh(

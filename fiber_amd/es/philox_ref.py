@@ -58,6 +58,17 @@ def normal4(k0, k1, c0, c1, c2, c3):
     return np.stack([z0, z1, z2, z3], axis=-1).astype(np.float32)
 
 
+def uniform4(k0, k1, c0, c1, c2, c3):
+    """4 uniforms in [-1, 1) per counter (matches fam_uniform4)."""
+    x0, x1, x2, x3 = philox4x32_10(k0, k1, c0, c1, c2, c3)
+    inv31 = np.float32(4.656612873077393e-10)
+    z = [
+        x.astype(np.int32).astype(np.float32) * inv31
+        for x in (x0, x1, x2, x3)
+    ]
+    return np.stack(z, axis=-1).astype(np.float32)
+
+
 def noise_for_pair(seed, iteration, pair, nparams):
     """The full eps vector (len nparams) for one antithetic pair."""
     nblocks = (nparams + 3) // 4

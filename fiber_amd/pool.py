@@ -232,6 +232,70 @@ def _execute_chunk(func, args, starmap, kwds):
 
 
 def _pool_worker_loop(
+    task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix,
+    nproc=1,
+):
+    """Worker job entry: runs ``nproc`` worker cores in this job
+    (reference multi-worker-per-job, ``cpu_per_job``: forked cores share
+    the shm rings; each core gets a derived ident for attribution).
+    Forking happens before any socket/HIP state exists in the core."""
+    if nproc > 1:
+        children = []
+        for k in range(1, nproc):
+            pid = os.fork()
+            if pid == 0:
+                _set_pdeathsig()  # die with the job's primary process
+                try:
+                    _pool_worker_core(task_addr, result_addr, resilient,
+                                      maxtasks, init_blob,
+                                      "%s:%d" % (ident_prefix, k))
+                except BaseException:
+                    # A core fault must take the whole job down so the
+                    # master's pending-table resubmission covers every
+                    # core's in-flight chunks (the job IS the failure
+                    # domain, reference cpu_per_job semantics).
+                    import signal as _signal
+
+                    os.kill(os.getppid(), _signal.SIGTERM)
+                    os._exit(1)
+                os._exit(0)
+            children.append(pid)
+        try:
+            _pool_worker_core(task_addr, result_addr, resilient, maxtasks,
+                              init_blob, ident_prefix)
+        except BaseException:
+            import signal as _signal
+
+            for pid in children:
+                try:
+                    os.kill(pid, _signal.SIGKILL)
+                except OSError:
+                    pass
+            raise
+        finally:
+            for pid in children:
+                try:
+                    os.waitpid(pid, 0)
+                except ChildProcessError:
+                    pass
+        return
+    _pool_worker_core(task_addr, result_addr, resilient, maxtasks,
+                      init_blob, ident_prefix)
+
+
+def _set_pdeathsig():
+    """Linux: deliver SIGKILL to this process when its parent dies."""
+    try:
+        import ctypes
+        import signal as _signal
+
+        libc = ctypes.CDLL("libc.so.6", use_errno=True)
+        libc.prctl(1, _signal.SIGKILL)  # PR_SET_PDEATHSIG
+    except Exception:
+        pass
+
+
+def _pool_worker_core(
     task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix
 ):
     """Worker main loop (reference zpool_worker_core, pool.py:760-825)."""
@@ -315,10 +379,12 @@ class ZPool:
         initargs=(),
         maxtasksperchild=None,
         gpu_per_worker=None,
+        cpu_per_worker=None,
         name=None,
     ):
         conf = fam_config.get_object()
         self._processes = processes or os.cpu_count() or 1
+        self._nproc_per_job = int(cpu_per_worker or conf.cpu_per_job or 1)
         self._maxtasks = maxtasksperchild
         self._name = name or util.random_name("fam-pool")
         self._gpu_per_worker = gpu_per_worker
@@ -387,6 +453,7 @@ class ZPool:
                 maxtasks=self._maxtasks,
                 init_blob=self._init_blob,
                 ident_prefix=ident,
+                nproc=self._nproc_per_job,
             ),
             self._meta,
         )
@@ -544,9 +611,9 @@ class ZPool:
         if self._state != "run":
             return
         self._state = "closing"
-        # One exit sentinel per worker rides the normal task channel (via
-        # the local queue) so it lands after all real tasks.
-        for _ in range(self._processes):
+        # One exit sentinel per worker core rides the normal task channel
+        # (via the local queue) so it lands after all real tasks.
+        for _ in range(self._processes * self._nproc_per_job):
             self._taskq.put((_SENTINEL_SEQ, 0, b"", [], False, None))
 
     def terminate(self):
@@ -589,13 +656,18 @@ class ZPool:
             self._worker_thread.join(timeout=2.0)
         self._task_sock.close()
         self._result_sock.close()
-        # Workers killed by SIGTERM never ran their atexit cleanup; reap
-        # their per-ident reply rings (resilient mode) master-side.
+        # Workers killed by SIGTERM/SIGKILL never ran their atexit
+        # cleanup; reap their per-ident reply rings (resilient mode)
+        # master-side, including forked cores' derived idents.
         for ident in self._all_idents:
-            try:
-                os.unlink("/dev/shm/%s.task.r.%s" % (self._name, ident))
-            except OSError:
-                pass
+            names = [ident] + [
+                "%s:%d" % (ident, k) for k in range(1, self._nproc_per_job)
+            ]
+            for name in names:
+                try:
+                    os.unlink("/dev/shm/%s.task.r.%s" % (self._name, name))
+                except OSError:
+                    pass
 
     def __enter__(self):
         return self
@@ -619,8 +691,9 @@ class ResilientZPool(ZPool):
     resilient = True
 
     def _alive(self, ident):
+        primary = ident.split(":", 1)[0]  # derived core idents share a job
         with self._worker_lock:
-            proc = self._workers.get(ident)
+            proc = self._workers.get(primary)
         return proc is not None and proc.exitcode is None
 
     def _send_task(self, task):
@@ -666,18 +739,23 @@ class ResilientZPool(ZPool):
             table.pop((seq, base), None)
 
     def _on_worker_death(self, ident, proc):
-        """Requeue everything the dead worker had claimed."""
-        table = self._pending.pop(ident, None)
-        self._task_sock.drop_peer(ident)
-        if table:
-            log = util.get_logger()
-            log.warning(
-                "worker %s died with %d pending chunks; resubmitting",
-                ident,
-                len(table),
-            )
-            for task in table.values():
-                self._taskq.put(task)
+        """Requeue everything the dead job's cores had claimed."""
+        dead_idents = [
+            k
+            for k in list(self._pending)
+            if k == ident or k.startswith(ident + ":")
+        ] or [ident]
+        for did in dead_idents:
+            table = self._pending.pop(did, None)
+            self._task_sock.drop_peer(did)
+            if table:
+                util.get_logger().warning(
+                    "worker %s died with %d pending chunks; resubmitting",
+                    did,
+                    len(table),
+                )
+                for task in table.values():
+                    self._taskq.put(task)
 
 
 Pool = ResilientZPool

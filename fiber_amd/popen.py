@@ -193,9 +193,14 @@ class Popen:
             code = self._backend.get_job_exitcode(self._job)
             if code is not None:
                 listener.take(ident)
+                logs = ""
+                try:
+                    logs = self._backend.get_job_logs(self._job)[-2000:]
+                except Exception:
+                    pass
                 raise RuntimeError(
-                    "worker %s exited with code %s before handshake"
-                    % (process_obj.name, code)
+                    "worker %s exited with code %s before handshake; "
+                    "logs:\n%s" % (process_obj.name, code, logs)
                 )
         self._conn = listener.take(ident)
 
@@ -219,13 +224,31 @@ class Popen:
     def poll(self):
         if self._exitcode is None:
             self._exitcode = self._backend.get_job_exitcode(self._job)
+            if self._exitcode is not None:
+                self._reap_log()
         return self._exitcode
+
+    def _reap_log(self):
+        """Remove the job's capture file once it exited cleanly & silent."""
+        path = getattr(self._job, "log_path", None)
+        if path and self._exitcode == 0:
+            try:
+                if os.path.getsize(path) == 0:
+                    os.unlink(path)
+                    self._job.log_path = None
+            except OSError:
+                pass
 
     def wait(self, timeout=None):
         if self._exitcode is not None:
             return self._exitcode
         self._exitcode = self._backend.wait_for_job(self._job, timeout)
+        if self._exitcode is not None:
+            self._reap_log()
         return self._exitcode
+
+    def get_logs(self):
+        return self._backend.get_job_logs(self._job)
 
     def terminate(self):
         self._backend.terminate_job(self._job)

@@ -157,6 +157,12 @@ class Process:
         if self._popen is not None:
             self._popen.kill()
 
+    def logs(self):
+        """Captured stdout/stderr of the job (reference get_job_logs)."""
+        if self._popen is None:
+            return ""
+        return self._popen.get_logs()
+
     def close(self):
         if self._popen is not None and self._popen.poll() is None:
             raise ValueError("cannot close a process while it is still running")

@@ -146,6 +146,35 @@ class TestResilientZPool:
         p2.join()
 
 
+def _pid_task(x):
+    import os
+
+    time.sleep(0.05)
+    return os.getpid()
+
+
+class TestMultiWorkerPerJob:
+    def test_cpu_per_worker_forks_cores(self):
+        """cpu_per_job analog (reference pool.py:861-878): one job hosts
+        several worker cores sharing the rings."""
+        pool = ZPool(processes=2, cpu_per_worker=2)
+        try:
+            pids = set(pool.map(_pid_task, range(32), chunksize=1))
+            assert len(pids) >= 3  # 2 jobs x 2 cores, allow one laggard
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_resilient_with_forked_cores(self):
+        pool = ResilientZPool(processes=2, cpu_per_worker=2)
+        try:
+            res = pool.map(_random_error, range(120), chunksize=2)
+            assert res == [x * 2 for x in range(120)]
+        finally:
+            pool.terminate()
+            pool.join()
+
+
 class TestPoolInitializer:
     def test_initializer_runs_in_workers(self):
         def init(v):
