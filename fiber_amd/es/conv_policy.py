@@ -183,6 +183,21 @@ class ConvESEngine:
             "env_steps": self.pop_total * cfg.envs_per_member * cfg.horizon,
         }
 
+    def state_dict(self):
+        return {
+            "theta": self.theta.cpu(),
+            "adam_m": self.adam_m.cpu(),
+            "adam_v": self.adam_v.cpu(),
+            "t_step": self.t_step,
+            "config": dataclasses.asdict(self.cfg),
+        }
+
+    def load_state_dict(self, state):
+        self.theta.copy_(state["theta"].to(self.device))
+        self.adam_m.copy_(state["adam_m"].to(self.device))
+        self.adam_v.copy_(state["adam_v"].to(self.device))
+        self.t_step = int(state["t_step"])
+
 
 # ---------------------------------------------------------------------------
 # fp32 torch reference (bf16 rounding at the kernel's rounding points) —

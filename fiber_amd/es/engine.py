@@ -146,6 +146,43 @@ class ESEngine:
             "env_steps": self.pop_total * cfg.envs_per_member * cfg.horizon,
         }
 
+    # -- checkpoint / resume ----------------------------------------------
+    # (the reference has no checkpointing — SURVEY §5; this is additive.)
+    def state_dict(self):
+        return {
+            "theta": self.theta.cpu(),
+            "adam_m": self.adam_m.cpu(),
+            "adam_v": self.adam_v.cpu(),
+            "t_step": self.t_step,
+            "obs_sum": self.obs_sum.cpu(),
+            "obs_sumsq": self.obs_sumsq.cpu(),
+            "obs_count": self.obs_count.cpu(),
+            "config": dataclasses.asdict(self.cfg),
+        }
+
+    def load_state_dict(self, state):
+        self.theta.copy_(state["theta"].to(self.device))
+        self.adam_m.copy_(state["adam_m"].to(self.device))
+        self.adam_v.copy_(state["adam_v"].to(self.device))
+        self.t_step = int(state["t_step"])
+        self.obs_sum.copy_(state["obs_sum"].to(self.device))
+        self.obs_sumsq.copy_(state["obs_sumsq"].to(self.device))
+        self.obs_count.copy_(state["obs_count"].to(self.device))
+        count = self.obs_count.clamp_min(1.0)
+        self.obs_mu = (self.obs_sum / count).contiguous()
+        var = self.obs_sumsq / count - self.obs_mu ** 2
+        self.obs_nu = var.clamp_min(1e-2).contiguous()
+
+    def save(self, path):
+        import torch as _torch
+
+        _torch.save(self.state_dict(), path)
+
+    def load(self, path):
+        import torch as _torch
+
+        self.load_state_dict(_torch.load(path, weights_only=False))
+
 
 # ---------------------------------------------------------------------------
 # fp32 torch reference of one rollout batch (mirrors the kernel exactly,

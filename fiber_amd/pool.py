@@ -567,6 +567,23 @@ class ZPool:
             self._taskq.put((seq, base, func_blob, chunk, starmap, kwds))
         return result
 
+    def stats(self):
+        """Observability counters (reference only had sent/recv flow
+        counters, fiber/pool.py:902-904; extended here)."""
+        with self._worker_lock:
+            alive = sum(
+                1 for p in self._workers.values() if p.exitcode is None
+            )
+        return {
+            "tasks_sent": self._sent,
+            "results_received": self._recv,
+            "in_flight": self._sent - self._recv,
+            "workers_alive": alive,
+            "workers_spawned_total": len(self._all_idents),
+            "task_ring_depth": self._task_sock.pending,
+            "state": self._state,
+        }
+
     # multiprocessing.Pool API --------------------------------------------
     def apply(self, func, args=(), kwds=None):
         return self.apply_async(func, args, kwds).get()
