@@ -29,12 +29,13 @@ extern "C" __global__ void mfma_gemm64_probe(const float*, const float*,
                                              float*);
 // conv policy pipeline (conv_kernels.hip)
 extern "C" __global__ void es_perturb(const float*, int, int, float,
-                                      uint32_t, uint32_t, int,
+                                      uint32_t, const uint32_t*, int,
                                       __hip_bfloat16*);
-extern "C" __global__ void conv_env_init(uint32_t, uint32_t, int, float*,
-                                         float*);
+extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
+                                         float*, float*);
 extern "C" __global__ void conv_obsgen(const float*, const float*, uint32_t,
-                                       uint32_t, uint32_t, __hip_bfloat16*);
+                                       const uint32_t*, uint32_t,
+                                       __hip_bfloat16*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
                                        const __hip_bfloat16*, int,
                                        __hip_bfloat16*);
@@ -117,34 +118,34 @@ static void launch_gemm64_probe(uintptr_t a, uintptr_t b, uintptr_t c,
 // ---- conv pipeline launchers ---------------------------------------------
 
 static void launch_perturb(uintptr_t theta, int nparams, int np_pad,
-                           double sigma, uint32_t seed, uint32_t iter,
+                           double sigma, uint32_t seed, uintptr_t iterp,
                            int member_offset, int pop, uintptr_t wpert,
                            uintptr_t stream) {
   const int bx = 64;  // grid-stride over param blocks
   hipLaunchKernelGGL(es_perturb, dim3(bx, pop), dim3(256), 0,
                      (hipStream_t)stream, (const float*)theta, nparams,
-                     np_pad, (float)sigma, seed, iter, member_offset,
-                     (__hip_bfloat16*)wpert);
+                     np_pad, (float)sigma, seed, (const uint32_t*)iterp,
+                     member_offset, (__hip_bfloat16*)wpert);
   check(hipGetLastError(), "es_perturb launch");
 }
 
-static void launch_conv_env_init(uint32_t seed, uint32_t iter, int nmembers,
-                                 uintptr_t state, uintptr_t racc,
-                                 uintptr_t stream) {
+static void launch_conv_env_init(uint32_t seed, uintptr_t iterp,
+                                 int nmembers, uintptr_t state,
+                                 uintptr_t racc, uintptr_t stream) {
   const int n = nmembers * 16;
   hipLaunchKernelGGL(conv_env_init, dim3((n + 255) / 256), dim3(256), 0,
-                     (hipStream_t)stream, seed, iter, nmembers,
-                     (float*)state, (float*)racc);
+                     (hipStream_t)stream, seed, (const uint32_t*)iterp,
+                     nmembers, (float*)state, (float*)racc);
   check(hipGetLastError(), "conv_env_init launch");
 }
 
 static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
-                               uint32_t seed, uint32_t iter, uint32_t t,
+                               uint32_t seed, uintptr_t iterp, uint32_t t,
                                int nenv_total, uintptr_t obs,
                                uintptr_t stream) {
   hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
                      (hipStream_t)stream, (const float*)state,
-                     (const float*)gtab, seed, iter, t,
+                     (const float*)gtab, seed, (const uint32_t*)iterp, t,
                      (__hip_bfloat16*)obs);
   check(hipGetLastError(), "conv_obsgen launch");
 }
@@ -209,13 +210,13 @@ PYBIND11_MODULE(_ops, m) {
   m.attr("CONV_ENVS") = 16;
   m.def("es_perturb", &launch_perturb, py::arg("theta"), py::arg("nparams"),
         py::arg("np_pad"), py::arg("sigma"), py::arg("seed"),
-        py::arg("iter"), py::arg("member_offset"), py::arg("pop"),
+        py::arg("iterp"), py::arg("member_offset"), py::arg("pop"),
         py::arg("wpert"), py::arg("stream"));
   m.def("conv_env_init", &launch_conv_env_init, py::arg("seed"),
-        py::arg("iter"), py::arg("nmembers"), py::arg("state"),
+        py::arg("iterp"), py::arg("nmembers"), py::arg("state"),
         py::arg("racc"), py::arg("stream"));
   m.def("conv_obsgen", &launch_conv_obsgen, py::arg("state"),
-        py::arg("gtab"), py::arg("seed"), py::arg("iter"), py::arg("t"),
+        py::arg("gtab"), py::arg("seed"), py::arg("iterp"), py::arg("t"),
         py::arg("nenv_total"), py::arg("obs"), py::arg("stream"));
   m.def("conv_forward", &launch_conv_forward, py::arg("wpert"),
         py::arg("obs"), py::arg("act1"), py::arg("act2"), py::arg("act3"),

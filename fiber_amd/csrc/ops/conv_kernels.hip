@@ -70,9 +70,11 @@ __device__ inline float fast_tanh_c(float x) {
 // ---------------------------------------------------------------------------
 extern "C" __global__ void es_perturb(const float* __restrict__ theta,
                                       int nparams, int np_pad, float sigma,
-                                      uint32_t seed, uint32_t iter,
+                                      uint32_t seed,
+                                      const uint32_t* __restrict__ iterp,
                                       int member_offset,
                                       __hip_bfloat16* __restrict__ wpert) {
+  const uint32_t iter = *iterp;
   const int member = member_offset + blockIdx.y;
   const uint32_t pair = (uint32_t)(member >> 1);
   const float sgn = (member & 1) ? -sigma : sigma;
@@ -94,10 +96,12 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
 // ---------------------------------------------------------------------------
 // conv_env_init: state[b][e][d] = 0.3*z  (member-independent, like MLP)
 // ---------------------------------------------------------------------------
-extern "C" __global__ void conv_env_init(uint32_t seed, uint32_t iter,
+extern "C" __global__ void conv_env_init(uint32_t seed,
+                                         const uint32_t* __restrict__ iterp,
                                          int nmembers,
                                          float* __restrict__ state,
                                          float* __restrict__ racc) {
+  const uint32_t iter = *iterp;
   const int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= nmembers * CENV) return;
   const int e = idx % CENV;
@@ -115,9 +119,11 @@ extern "C" __global__ void conv_env_init(uint32_t seed, uint32_t iter,
 // ---------------------------------------------------------------------------
 extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
                                        const float* __restrict__ gtab,
-                                       uint32_t seed, uint32_t iter,
+                                       uint32_t seed,
+                                       const uint32_t* __restrict__ iterp,
                                        uint32_t t,
                                        __hip_bfloat16* __restrict__ obs) {
+  const uint32_t iter = *iterp;
   const int be = blockIdx.x;  // member*CENV + env
   const int e = be % CENV;
   __hip_bfloat16* out = obs + (size_t)be * (IMG * IMG * CIN);

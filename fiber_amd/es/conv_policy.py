@@ -115,25 +115,31 @@ class ConvESEngine:
                                  device=device)
         self.racc = torch.empty(pop, E, dtype=torch.float32, device=device)
         self._fitness_all = torch.empty(self.pop_total, device=device)
+        # iteration counter lives in device memory so a captured hipGraph
+        # of the whole rollout stays valid across ES iterations
+        self._iter_buf = torch.zeros(1, dtype=torch.int32, device=device)
+        self._graph = None
+        self.use_graph = True
 
     def _stream(self):
         return torch.cuda.current_stream().cuda_stream
 
-    def rollout(self, iteration):
-        """Run the shard's rollouts; returns fitness[pop] on device."""
+    def _rollout_body(self):
+        """The full rollout kernel DAG (captured into a hipGraph)."""
         o = ops._require_ops()
         cfg = self.cfg
         pop = cfg.pop_per_gpu
         member_offset = self.rank * pop
         stream = self._stream()
+        iterp = self._iter_buf.data_ptr()
         o.es_perturb(self.theta.data_ptr(), NP_CONV, o.NP_CONV_PAD,
-                     cfg.sigma, cfg.seed, iteration, member_offset, pop,
+                     cfg.sigma, cfg.seed, iterp, member_offset, pop,
                      self.wpert.data_ptr(), stream)
-        o.conv_env_init(cfg.seed, iteration, pop, self.state.data_ptr(),
+        o.conv_env_init(cfg.seed, iterp, pop, self.state.data_ptr(),
                         self.racc.data_ptr(), stream)
         for t in range(cfg.horizon):
             o.conv_obsgen(self.state.data_ptr(), self.gtab.data_ptr(),
-                          cfg.seed, iteration, t, pop * cfg.envs_per_member,
+                          cfg.seed, iterp, t, pop * cfg.envs_per_member,
                           self.obs.data_ptr(), stream)
             o.conv_forward(self.wpert.data_ptr(), self.obs.data_ptr(),
                            self.act1.data_ptr(), self.act2.data_ptr(),
@@ -142,6 +148,40 @@ class ConvESEngine:
                             pop, self.env_A.data_ptr(),
                             self.env_B.data_ptr(), self.state.data_ptr(),
                             self.racc.data_ptr(), stream)
+
+    def _ensure_graph(self):
+        if self._graph is not None:
+            return True
+        if not self.use_graph:
+            return False
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._rollout_body()  # warmup outside capture
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                self._rollout_body()
+            self._graph = graph
+            return True
+        except Exception:
+            self.use_graph = False
+            return False
+
+    def rollout(self, iteration):
+        """Run the shard's rollouts; returns fitness[pop] on device.
+
+        The T-step 5-kernel pipeline is replayed as ONE hipGraph (the
+        iteration counter is read from device memory), eliminating
+        horizon x 5 launch + host-loop overheads; falls back to eager
+        launches if graph capture is unavailable."""
+        self._iter_buf.fill_(int(iteration))
+        if self._ensure_graph():
+            self._graph.replay()
+        else:
+            self._rollout_body()
         return self.racc.mean(dim=1)
 
     def step(self, iteration=None):

@@ -30,8 +30,9 @@ class TestConvPipeline:
         pop = 4
         wpert = torch.empty(pop, o.NP_CONV_PAD, dtype=torch.bfloat16,
                             device=device)
+        iterp = torch.zeros(1, dtype=torch.int32, device=device)
         o.es_perturb(theta.data_ptr(), conv_policy.NP_CONV, o.NP_CONV_PAD,
-                     0.02, 11, 0, 0, pop, wpert.data_ptr(),
+                     0.02, 11, iterp.data_ptr(), 0, pop, wpert.data_ptr(),
                      torch.cuda.current_stream().cuda_stream)
         torch.cuda.synchronize()
         for m in (0, 1, 3):
