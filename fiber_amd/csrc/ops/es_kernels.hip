@@ -161,9 +161,7 @@ struct RolloutLds {
   };
   alignas(16) __hip_bfloat16 h1[ENVS][S2];  // B-operand layer 2
   float S[2][ENVS][OBS];      // env state (double-buffered)
-  float lpart[OBS][ENVS][2];  // logits partials (phase D1 -> D2)
-  float sq[OBS][ENVS];        // squared-state partials (D2 -> next A)
-  float racc[ENVS];           // per-env return
+  float lpart[OBS][ENVS][2];  // logits partials (D1 -> D2; epilogue reuse)
   float ostat[2 * OBS + 1];   // sum, sumsq, count
 };
 
@@ -211,41 +209,39 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     fam_normal4(seed, iter, (uint32_t)tid, 0u, FAM_TAG_ENV, 0u, z);
 #pragma unroll
     for (int d = 0; d < OBS; ++d) L.S[0][tid][d] = 0.3f * z[d];
-    L.racc[tid] = 0.f;
   }
   if (tid < 2 * OBS + 1) L.ostat[tid] = 0.f;
   __syncthreads();
 
   // Every phase below uses all 256 threads as (env = tid&63, d = tid>>6)
-  // — OBS==4 matches the 4 waves exactly.  Per-thread scalar obs-stat
-  // partials; reward partials go through L.part to stay deterministic
-  // (no float atomics on the fitness path).
+  // — OBS==4 matches the 4 waves exactly.  The thread that owns (env, d)
+  // keeps its state component, obs-stat partials AND reward partial in
+  // REGISTERS across the whole horizon: phase A is folded into the tail
+  // of D2 (4 barriers per step), the fitness path stays deterministic
+  // (register partials, no float atomics), and only the cross-dim env
+  // state round-trips through LDS (for D2's drive term).
   const int env = tid & 63;
   const int dd = tid >> 6;  // 0..3
-  float psum = 0.f, psq = 0.f;
+  float psum = 0.f, psq = 0.f, raccp = 0.f;
   const float mu_d = obs_mu[dd];
   const float rstd_d = rsqrtf(obs_nu[dd] + 1e-4f);
   float eA_d[OBS];
 #pragma unroll
   for (int e = 0; e < OBS; ++e) eA_d[e] = env_A[dd * OBS + e];
   const float eB_d = env_B[dd];
+  const float one_if_d0 = (dd == 0) ? 1.f : 0.f;
+
+  // prologue "phase A" for t=0: stats + normalized obs from the init state
+  {
+    const float s = L.S[0][env][dd];
+    psum += s;
+    psq += s * s;
+    float x = (s - mu_d) * rstd_d;
+    L.xb[env][dd] = __float2bfloat16(fminf(5.f, fmaxf(-5.f, x)));
+  }
 
   for (int t = 0; t < horizon; ++t) {
     const int cur = t & 1;
-    // ---- phase A: normalize obs -> xb; stats; fold reward partials ----
-    {
-      const float s = L.S[cur][env][dd];
-      psum += s;
-      psq += s * s;
-      float x = (s - mu_d) * rstd_d;
-      x = fminf(5.f, fmaxf(-5.f, x));
-      L.xb[env][dd] = __float2bfloat16(x);
-      if (dd == 0 && t > 0) {
-        // reward for step t-1 (squared-state partials staged in L.sq)
-        L.racc[env] += 1.f - 0.1f * (L.sq[0][env] + L.sq[1][env] +
-                                     L.sq[2][env] + L.sq[3][env]);
-      }
-    }
     __syncthreads();
     // ---- phase B: h1 = tanh(W1 x + b1)  (MFMA, K=32) ------------------
     mfma_strip_tanh<KPAD, S1, S1, S2>(&L.pol.w1[0][0], &L.xb[0][0],
@@ -269,7 +265,7 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
       L.lpart[dd][env][1] = p1;
     }
     __syncthreads();
-    // ---- phase D2: action + env step (thread (env, d)) ----------------
+    // ---- phase D2 (+A of t+1): action, env step, stats, next xb -------
     {
       const float l0 = L.pol.b3[0] + L.lpart[0][env][0] +
                        L.lpart[1][env][0] + L.lpart[2][env][0] +
@@ -284,21 +280,28 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
       const float snew = 0.97f * L.S[cur][env][dd] +
                          0.08f * fast_tanh(drive) + 0.05f * eB_d * asign;
       L.S[cur ^ 1][env][dd] = snew;
-      L.sq[dd][env] = snew * snew;  // staged for next phase-A reward
+      raccp += one_if_d0 - 0.1f * snew * snew;
+      if (t < horizon - 1) {
+        // next step's stats + normalized obs, from the register state
+        psum += snew;
+        psq += snew * snew;
+        float x = (snew - mu_d) * rstd_d;
+        L.xb[env][dd] = __float2bfloat16(fminf(5.f, fmaxf(-5.f, x)));
+      }
     }
-    __syncthreads();
   }
+  __syncthreads();
 
-  // ---- epilogue: last step's reward, fitness, obs-stat reduction -------
-  if (tid < ENVS) {
-    L.racc[tid] += 1.f - 0.1f * (L.sq[0][tid] + L.sq[1][tid] +
-                                 L.sq[2][tid] + L.sq[3][tid]);
-  }
+  // ---- epilogue: fitness + obs-stat reductions -------------------------
+  // reward partials: 4 per env (one per dim-owner thread), staged via
+  // lpart then wave-0 shuffle-reduced — deterministic order.
+  L.lpart[dd][env][0] = raccp;
   atomicAdd(&L.ostat[dd], psum);
   atomicAdd(&L.ostat[OBS + dd], psq);
   __syncthreads();
   if (wave == 0) {
-    float r = L.racc[lane];
+    float r = L.lpart[0][lane][0] + L.lpart[1][lane][0] +
+              L.lpart[2][lane][0] + L.lpart[3][lane][0];
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) r += __shfl_down(r, off, 64);
     if (lane == 0) fitness[blockIdx.x] = r / (float)ENVS;
