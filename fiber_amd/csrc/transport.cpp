@@ -31,6 +31,7 @@
 #include <pthread.h>
 #include <sys/mman.h>
 #include <sys/stat.h>
+#include <signal.h>
 #include <time.h>
 #include <unistd.h>
 
@@ -41,6 +42,14 @@ namespace {
 constexpr uint32_t kMagic = 0xFA3B71A6u;
 constexpr uint32_t kWrapMarker = 0xFFFFFFFFu;
 constexpr size_t kAlign = 8;
+
+// Record layout: [u32 len][u32 state][u32 writer_pid][u32 _pad][payload..]
+// state: 0 = reserved (payload being copied outside the lock),
+//        1 = committed.  A reserved record whose writer died is reclaimed
+// by the next reader (crash safety for mid-copy kills).
+constexpr uint32_t kStReserved = 0u;
+constexpr uint32_t kStCommitted = 1u;
+constexpr size_t kRecHdr = 16;
 
 struct Header {
   uint32_t magic;
@@ -59,7 +68,12 @@ struct Header {
 };
 
 inline size_t record_bytes(size_t len) {
-  return (4 + len + kAlign - 1) & ~(kAlign - 1);
+  return (kRecHdr + len + kAlign - 1) & ~(kAlign - 1);
+}
+
+inline bool process_alive(uint32_t pid) {
+  if (pid == 0) return false;
+  return kill((pid_t)pid, 0) == 0 || errno != ESRCH;
 }
 
 inline void make_deadline(double timeout_s, struct timespec* ts) {
@@ -198,46 +212,65 @@ class ShmRing {
 
   // timeout < 0: block forever; timeout == 0: non-blocking.
   // Returns false on timeout; throws if the ring is closed.
+  // The payload memcpy happens OUTSIDE the lock (reserve -> copy ->
+  // commit), so concurrent producers/consumers overlap their copies.
   bool send(const char* buf, size_t len, double timeout) {
     size_t need = record_bytes(len);
-    if (need + kAlign > hdr_->capacity)
+    if (need + kRecHdr + kAlign >= hdr_->capacity)
       throw std::runtime_error("message larger than ring capacity");
     struct timespec deadline;
     if (timeout > 0) make_deadline(timeout, &deadline);
 
-    RobustLock lock(&hdr_->mu);
-    for (;;) {
-      if (hdr_->closed) throw std::runtime_error("ring closed");
-      // Worst case we also need a wrap marker record.
-      if (hdr_->capacity - hdr_->used >= need + 4 + kAlign) break;
-      if (timeout == 0) return false;
-      int rc;
-      if (timeout < 0) {
-        rc = pthread_cond_wait(&hdr_->not_full, &hdr_->mu);
-      } else {
-        rc = pthread_cond_timedwait(&hdr_->not_full, &hdr_->mu, &deadline);
-        if (rc == ETIMEDOUT) return false;
+    uint64_t rec;
+    {
+      RobustLock lock(&hdr_->mu);
+      for (;;) {
+        if (hdr_->closed) throw std::runtime_error("ring closed");
+        // Worst case we also need a wrap marker record.
+        if (hdr_->capacity - hdr_->used >= need + kRecHdr + kAlign) break;
+        if (timeout == 0) return false;
+        int rc;
+        if (timeout < 0) {
+          rc = pthread_cond_wait(&hdr_->not_full, &hdr_->mu);
+        } else {
+          rc = pthread_cond_timedwait(&hdr_->not_full, &hdr_->mu,
+                                      &deadline);
+          if (rc == ETIMEDOUT) return false;
+        }
+        (void)rc;
       }
-      (void)rc;
+
+      uint64_t cap = hdr_->capacity;
+      uint64_t tail = hdr_->tail;
+      if (tail + need > cap) {
+        // Not enough contiguous space: write wrap marker, jump to 0.
+        uint32_t marker = kWrapMarker;
+        std::memcpy(data_ + tail, &marker, 4);
+        hdr_->used += cap - tail;
+        tail = 0;
+      }
+      rec = tail;
+      uint32_t len32 = (uint32_t)len;
+      uint32_t pid = (uint32_t)getpid();
+      std::memcpy(data_ + rec, &len32, 4);
+      reinterpret_cast<std::atomic<uint32_t>*>(data_ + rec + 4)
+          ->store(kStReserved, std::memory_order_relaxed);
+      std::memcpy(data_ + rec + 8, &pid, 4);
+      hdr_->tail = (tail + need) % cap;
+      hdr_->used += need;
+      hdr_->msg_count += 1;
+      hdr_->total_in += 1;
     }
 
-    uint64_t cap = hdr_->capacity;
-    uint64_t tail = hdr_->tail;
-    if (tail + need > cap) {
-      // Not enough contiguous space: write wrap marker, jump to 0.
-      uint32_t marker = kWrapMarker;
-      std::memcpy(data_ + tail, &marker, 4);
-      hdr_->used += cap - tail;
-      tail = 0;
+    if (len) std::memcpy(data_ + rec + kRecHdr, buf, len);
+    reinterpret_cast<std::atomic<uint32_t>*>(data_ + rec + 4)
+        ->store(kStCommitted, std::memory_order_release);
+    {
+      // Lock-protected signal so a reader between predicate-check and
+      // cond_wait cannot miss the wakeup.
+      RobustLock lock(&hdr_->mu);
+      pthread_cond_signal(&hdr_->not_empty);
     }
-    uint32_t len32 = (uint32_t)len;
-    std::memcpy(data_ + tail, &len32, 4);
-    if (len) std::memcpy(data_ + tail + 4, buf, len);
-    hdr_->tail = (tail + need) % cap;
-    hdr_->used += need;
-    hdr_->msg_count += 1;
-    hdr_->total_in += 1;
-    pthread_cond_signal(&hdr_->not_empty);
     return true;
   }
 
@@ -246,10 +279,70 @@ class ShmRing {
   bool recv(std::string* out, double timeout) {
     struct timespec deadline;
     if (timeout > 0) make_deadline(timeout, &deadline);
+    // Bounded wait even for infinite timeouts so dead-writer reclaim
+    // gets a chance to run.
+    struct timespec tick;
 
     RobustLock lock(&hdr_->mu);
+    int stuck_polls = 0;
     for (;;) {
-      if (hdr_->msg_count > 0) break;
+      if (hdr_->msg_count > 0) {
+        // resolve head (following a wrap marker) and check commit state
+        uint64_t cap = hdr_->capacity;
+        uint64_t head = hdr_->head;
+        uint32_t len32;
+        std::memcpy(&len32, data_ + head, 4);
+        if (len32 == kWrapMarker) {
+          hdr_->used -= cap - head;
+          hdr_->head = head = 0;
+          std::memcpy(&len32, data_ + head, 4);
+        }
+        uint32_t state =
+            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
+                ->load(std::memory_order_acquire);
+        if (state == kStCommitted) {
+          out->assign(data_ + head + kRecHdr, len32);
+          size_t need = record_bytes(len32);
+          hdr_->head = (head + need) % cap;
+          hdr_->used -= need;
+          hdr_->msg_count -= 1;
+          hdr_->total_out += 1;
+          pthread_cond_signal(&hdr_->not_full);
+          return true;
+        }
+        // Reserved record: writer is copying.  If the writer died
+        // mid-copy, reclaim the record so the ring cannot wedge.
+        if (stuck_polls > 20) {  // ~2s of 100ms polls
+          uint32_t pid;
+          std::memcpy(&pid, data_ + head + 8, 4);
+          if (!process_alive(pid)) {
+            size_t need = record_bytes(len32);
+            hdr_->head = (head + need) % cap;
+            hdr_->used -= need;
+            hdr_->msg_count -= 1;
+            pthread_cond_signal(&hdr_->not_full);
+            stuck_polls = 0;
+            continue;
+          }
+          stuck_polls = 0;
+        }
+        // short poll wait for the commit
+        make_deadline(0.1, &tick);
+        int rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        if (rc == ETIMEDOUT) {
+          ++stuck_polls;
+          if (timeout > 0) {
+            struct timespec now;
+            clock_gettime(CLOCK_MONOTONIC, &now);
+            if (now.tv_sec > deadline.tv_sec ||
+                (now.tv_sec == deadline.tv_sec &&
+                 now.tv_nsec >= deadline.tv_nsec))
+              return false;
+          }
+          if (timeout == 0) return false;
+        }
+        continue;
+      }
       if (hdr_->closed) throw std::runtime_error("ring closed");
       if (timeout == 0) return false;
       int rc;
@@ -261,24 +354,123 @@ class ShmRing {
       }
       (void)rc;
     }
+  }
 
-    uint64_t cap = hdr_->capacity;
-    uint64_t head = hdr_->head;
-    uint32_t len32;
-    std::memcpy(&len32, data_ + head, 4);
-    if (len32 == kWrapMarker) {
-      hdr_->used -= cap - head;
-      head = 0;
-      std::memcpy(&len32, data_ + head, 4);
+  // Size of the next committed message, or -1 on timeout.  Does not
+  // consume.  (Used with recv_into for single-copy receives.)
+  int64_t peek_size(double timeout) {
+    struct timespec deadline;
+    if (timeout > 0) make_deadline(timeout, &deadline);
+    struct timespec tick;
+    RobustLock lock(&hdr_->mu);
+    for (;;) {
+      if (hdr_->msg_count > 0) {
+        uint64_t cap = hdr_->capacity;
+        uint64_t head = hdr_->head;
+        uint32_t len32;
+        std::memcpy(&len32, data_ + head, 4);
+        if (len32 == kWrapMarker) {
+          hdr_->used -= cap - head;
+          hdr_->head = head = 0;
+          std::memcpy(&len32, data_ + head, 4);
+        }
+        uint32_t state =
+            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
+                ->load(std::memory_order_acquire);
+        if (state == kStCommitted) return (int64_t)len32;
+        make_deadline(0.1, &tick);
+        pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        if (timeout == 0) return -1;
+        continue;
+      }
+      if (hdr_->closed) throw std::runtime_error("ring closed");
+      if (timeout == 0) return -1;
+      int rc;
+      if (timeout < 0) {
+        rc = pthread_cond_wait(&hdr_->not_empty, &hdr_->mu);
+      } else {
+        rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &deadline);
+        if (rc == ETIMEDOUT) return -1;
+      }
+      (void)rc;
     }
-    out->assign(data_ + head + 4, len32);
-    size_t need = record_bytes(len32);
-    hdr_->head = (head + need) % cap;
-    hdr_->used -= need;
-    hdr_->msg_count -= 1;
-    hdr_->total_out += 1;
-    pthread_cond_signal(&hdr_->not_full);
-    return true;
+  }
+
+  // Single-copy receive: copies the next committed message into buf.
+  // Returns the message length, -1 on timeout, or -(len) - 2 if buf is
+  // too small (message left in the ring).
+  int64_t recv_into(char* buf, size_t buflen, double timeout) {
+    struct timespec deadline;
+    if (timeout > 0) make_deadline(timeout, &deadline);
+    struct timespec tick;
+    RobustLock lock(&hdr_->mu);
+    int stuck_polls = 0;
+    for (;;) {
+      if (hdr_->msg_count > 0) {
+        uint64_t cap = hdr_->capacity;
+        uint64_t head = hdr_->head;
+        uint32_t len32;
+        std::memcpy(&len32, data_ + head, 4);
+        if (len32 == kWrapMarker) {
+          hdr_->used -= cap - head;
+          hdr_->head = head = 0;
+          std::memcpy(&len32, data_ + head, 4);
+        }
+        uint32_t state =
+            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
+                ->load(std::memory_order_acquire);
+        if (state == kStCommitted) {
+          if ((size_t)len32 > buflen) return -((int64_t)len32) - 2;
+          std::memcpy(buf, data_ + head + kRecHdr, len32);
+          size_t need = record_bytes(len32);
+          hdr_->head = (head + need) % cap;
+          hdr_->used -= need;
+          hdr_->msg_count -= 1;
+          hdr_->total_out += 1;
+          pthread_cond_signal(&hdr_->not_full);
+          return (int64_t)len32;
+        }
+        if (stuck_polls > 20) {
+          uint32_t pid;
+          std::memcpy(&pid, data_ + head + 8, 4);
+          if (!process_alive(pid)) {
+            size_t need = record_bytes(len32);
+            hdr_->head = (head + need) % cap;
+            hdr_->used -= need;
+            hdr_->msg_count -= 1;
+            pthread_cond_signal(&hdr_->not_full);
+            stuck_polls = 0;
+            continue;
+          }
+          stuck_polls = 0;
+        }
+        make_deadline(0.1, &tick);
+        int rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        if (rc == ETIMEDOUT) {
+          ++stuck_polls;
+          if (timeout == 0) return -1;
+          if (timeout > 0) {
+            struct timespec now;
+            clock_gettime(CLOCK_MONOTONIC, &now);
+            if (now.tv_sec > deadline.tv_sec ||
+                (now.tv_sec == deadline.tv_sec &&
+                 now.tv_nsec >= deadline.tv_nsec))
+              return -1;
+          }
+        }
+        continue;
+      }
+      if (hdr_->closed) throw std::runtime_error("ring closed");
+      if (timeout == 0) return -1;
+      int rc;
+      if (timeout < 0) {
+        rc = pthread_cond_wait(&hdr_->not_empty, &hdr_->mu);
+      } else {
+        rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &deadline);
+        if (rc == ETIMEDOUT) return -1;
+      }
+      (void)rc;
+    }
   }
 
   void close_ring() {
@@ -344,6 +536,27 @@ PYBIND11_MODULE(_transport, m) {
             return py::bytes(out);
           },
           py::arg("timeout") = -1.0)
+      .def(
+          "peek_size",
+          [](ShmRing& r, double timeout) {
+            py::gil_scoped_release release;
+            return r.peek_size(timeout);
+          },
+          py::arg("timeout") = -1.0)
+      .def(
+          "recv_into",
+          [](ShmRing& r, py::buffer buf, double timeout) {
+            py::buffer_info info = buf.request(true);
+            char* ptr = static_cast<char*>(info.ptr);
+            size_t cap = (size_t)info.size * (size_t)info.itemsize;
+            int64_t n;
+            {
+              py::gil_scoped_release release;
+              n = r.recv_into(ptr, cap, timeout);
+            }
+            return n;
+          },
+          py::arg("buf"), py::arg("timeout") = -1.0)
       .def("close", &ShmRing::close_ring,
            py::call_guard<py::gil_scoped_release>())
       .def("unlink", &ShmRing::unlink_ring)

@@ -318,9 +318,9 @@ def _pool_worker_core(
             task_sock.send(b"", timeout=-1.0)
             # Timeout + re-request guards against a discarded request (the
             # master drops requests it cannot attribute to a live worker).
-            payload = task_sock.recv(timeout=10.0)
+            payload = task_sock.recv_view(timeout=10.0)
         else:
-            payload = task_sock.recv(timeout=-1.0)
+            payload = task_sock.recv_view(timeout=-1.0)
         if payload is None:
             continue
         task = serialization.loads(payload)
@@ -521,9 +521,13 @@ class ZPool:
         self._task_sock.send(serialization.dumps(task), timeout=-1.0)
 
     def _result_loop(self):
+        from .transport import ring_recv_view
+
         while True:
             try:
-                payload = self._result_sock._rings["main"].recv(0.2)
+                payload = ring_recv_view(
+                    self._result_sock._rings["main"], 0.2
+                )
             except RuntimeError:
                 return
             except Exception:

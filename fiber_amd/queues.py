@@ -17,7 +17,7 @@ tensors put on a queue cross as HIP IPC handles, never by value.
 import queue as _stdlib_queue
 
 from . import serialization
-from .transport import Socket, new_address
+from .transport import Socket, new_address, ring_recv_view as _ring_recv
 
 
 class Connection:
@@ -46,7 +46,7 @@ class Connection:
         if self._buffered is not None:
             data, self._buffered = self._buffered, None
             return serialization.loads(data)
-        data = self._ensure().recv(timeout)
+        data = self._ensure().recv_view(timeout)
         if data is None:
             raise TimeoutError("recv timed out")
         return serialization.loads(data)
@@ -149,14 +149,14 @@ class SimpleQueue:
 
     def get(self, timeout=-1.0):
         ring = self._ensure()
-        data = ring.recv(timeout)
+        data = _ring_recv(ring, timeout)
         if data is None:
             raise TimeoutError("queue get timed out")
         return serialization.loads(data)
 
     def get_nowait(self):
         ring = self._ensure()
-        data = ring.recv(0.0)
+        data = _ring_recv(ring, 0.0)
         if data is None:
             raise _stdlib_queue.Empty
         return serialization.loads(data)

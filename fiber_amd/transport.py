@@ -59,6 +59,31 @@ def new_address(prefix="fam-ch"):
     return util.random_name(prefix)
 
 
+class _RecvBuffers(threading.local):
+    """Per-thread reusable receive buffer for single-copy ring reads."""
+
+    def __init__(self):
+        self.buf = bytearray(64 << 10)
+
+
+_recv_buffers = _RecvBuffers()
+
+
+def ring_recv_view(ring, timeout):
+    """Single-copy receive: returns a memoryview into a thread-local
+    buffer (valid until the next recv on this thread), or None on
+    timeout."""
+    buf = _recv_buffers.buf
+    while True:
+        n = ring.recv_into(buf, timeout)
+        if n >= 0:
+            return memoryview(buf)[:n]
+        if n == -1:
+            return None
+        need = -(n + 2)  # buffer too small: message is `need` bytes
+        _recv_buffers.buf = buf = bytearray(max(need, 2 * len(buf)))
+
+
 def _capacity():
     return fam_config.get_object().ring_capacity
 
@@ -114,15 +139,22 @@ class Socket:
             return self._rings["q"].send(envelope, timeout)
         raise ValueError("socket mode %r cannot send()" % self.mode)
 
-    def recv(self, timeout=-1.0):
+    def _recv_ring(self):
         if self.mode == "r":
-            return self._rings["main"].recv(timeout)
+            return self._rings["main"]
         if self.mode == "rw":
-            ring = self._rings["a"] if not self.bound else self._rings["b"]
-            return ring.recv(timeout)
+            return self._rings["a"] if not self.bound else self._rings["b"]
         if self.mode == "req":
-            return self._rings["r"].recv(timeout)
+            return self._rings["r"]
         raise ValueError("socket mode %r cannot recv()" % self.mode)
+
+    def recv(self, timeout=-1.0):
+        return self._recv_ring().recv(timeout)
+
+    def recv_view(self, timeout=-1.0):
+        """Single-copy receive into a thread-local buffer (memoryview
+        valid until this thread's next recv_view)."""
+        return ring_recv_view(self._recv_ring(), timeout)
 
     # -- rep ---------------------------------------------------------------
     def recv_request(self, timeout=-1.0):
