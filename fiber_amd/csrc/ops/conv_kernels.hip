@@ -24,12 +24,14 @@
 // step, amortized over all 16 envs).
 
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 
 #include "philox.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef long fp8x8;  // 8 packed OCP e4m3 bytes (MFMA fp8 operand)
 
 #define FAM_TAG_NOISE 0x45530001u
 #define FAM_TAG_ENV 0x45530002u
@@ -73,12 +75,15 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
                                       uint32_t seed,
                                       const uint32_t* __restrict__ iterp,
                                       int member_offset,
-                                      __hip_bfloat16* __restrict__ wpert) {
+                                      __hip_bfloat16* __restrict__ wpert,
+                                      unsigned char* __restrict__ w3_fp8) {
   const uint32_t iter = *iterp;
   const int member = member_offset + blockIdx.y;
   const uint32_t pair = (uint32_t)(member >> 1);
   const float sgn = (member & 1) ? -sigma : sigma;
   __hip_bfloat16* out = wpert + (size_t)blockIdx.y * np_pad;
+  unsigned char* out8 =
+      w3_fp8 ? w3_fp8 + (size_t)blockIdx.y * (FCU * NFLAT) : nullptr;
   const int jb0 = blockIdx.x * blockDim.x + threadIdx.x;
   const int stride = gridDim.x * blockDim.x;
   for (int jb = jb0; jb * 4 < nparams; jb += stride) {
@@ -88,7 +93,15 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
       const int j = j0 + u;
-      if (j < nparams) out[j] = __float2bfloat16(theta[j] + sgn * z[u]);
+      if (j < nparams) {
+        const float v = theta[j] + sgn * z[u];
+        out[j] = __float2bfloat16(v);
+        // the fc layer consumes W3 in OCP e4m3 (fp8) — halves the
+        // dominant HBM term of the rollout (profiles: conv_fc is
+        // BW-bound on single-use weights)
+        if (out8 && j >= COFF_W3 && j < COFF_B3)
+          out8[j - COFF_W3] = __hip_fp8_e4m3(v).__x;
+      }
     }
   }
 }
@@ -211,14 +224,14 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
 extern "C" __global__ void __launch_bounds__(256)
 conv_layer2(const __hip_bfloat16* __restrict__ wpert,
             const __hip_bfloat16* __restrict__ act1, int nenv_total,
-            __hip_bfloat16* __restrict__ act2) {
+            unsigned char* __restrict__ act2) {
   __shared__ alignas(16) __hip_bfloat16 w2[C2][256];
   __shared__ float b2[C2];
   const int be = blockIdx.x;
   const int member = be / CENV;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
   const __hip_bfloat16* in = act1 + (size_t)be * (O1 * O1 * C1);
-  __hip_bfloat16* out = act2 + (size_t)be * NFLAT;
+  unsigned char* out = act2 + (size_t)be * NFLAT;
 
   const int tid = threadIdx.x;
   for (int i = tid; i < C2 * 256 / 8; i += blockDim.x) {
@@ -258,15 +271,14 @@ conv_layer2(const __hip_bfloat16* __restrict__ wpert,
     const int dpos = nt * 16 + (lane & 15);
     if (dpos < O2 * O2) {
       union {
-        __hip_bfloat16 h[4];
-        unsigned long long u;
+        unsigned char b[4];
+        uint32_t u;
       } pk;
 #pragma unroll
       for (int ri = 0; ri < 4; ++ri)
-        pk.h[ri] = __float2bfloat16(
-            fast_tanh_c(acc[ri] + b2[(drow + ri) & 31]));
-      *reinterpret_cast<unsigned long long*>(
-          &out[dpos * C2 + (drow & 31)]) = pk.u;
+        pk.b[ri] = __hip_fp8_e4m3(
+            fast_tanh_c(acc[ri] + b2[(drow + ri) & 31])).__x;
+      *reinterpret_cast<uint32_t*>(&out[dpos * C2 + (drow & 31)]) = pk.u;
     }
   }
 }
@@ -278,12 +290,13 @@ conv_layer2(const __hip_bfloat16* __restrict__ wpert,
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(256)
 conv_fc(const __hip_bfloat16* __restrict__ wpert,
-        const __hip_bfloat16* __restrict__ act2, int nmembers,
+        const unsigned char* __restrict__ w3_fp8,
+        const unsigned char* __restrict__ act2, int nmembers,
         __hip_bfloat16* __restrict__ act3) {
   const int member = blockIdx.x;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
-  const __hip_bfloat16* w3 = wm + COFF_W3;
-  const __hip_bfloat16* in = act2 + (size_t)member * CENV * NFLAT;
+  const unsigned char* w3 = w3_fp8 + (size_t)member * (FCU * NFLAT);
+  const unsigned char* in = act2 + (size_t)member * CENV * NFLAT;
   __hip_bfloat16* out = act3 + (size_t)member * CENV * FCU;
 
   const int tid = threadIdx.x;
@@ -296,12 +309,12 @@ conv_fc(const __hip_bfloat16* __restrict__ wpert,
     const int arow = mt * 16 + (lane & 15);
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < 81; ++kk) {
-      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+      const fp8x8 afrag = *reinterpret_cast<const fp8x8*>(
           &w3[(size_t)arow * NFLAT + kk * 32 + kgrp * 8]);
-      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+      const fp8x8 bfrag = *reinterpret_cast<const fp8x8*>(
           &in[(size_t)env * NFLAT + kk * 32 + kgrp * 8]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0,
-                                                    0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag, bfrag, acc,
+                                                       0, 0, 0);
     }
     const int drow = mt * 16 + kgrp * 4;
     union {

@@ -106,10 +106,15 @@ class ConvESEngine:
 
         bf = torch.bfloat16
         self.wpert = torch.empty(pop, o.NP_CONV_PAD, dtype=bf, device=device)
+        # fc weights + activations ride OCP e4m3 (fp8): the fc layer is
+        # HBM-bound on single-use weights, fp8 halves the traffic
+        self.w3_fp8 = torch.empty(pop, 256 * 2592, dtype=torch.uint8,
+                                  device=device)
         self.obs = torch.empty(pop * E, 84 * 84 * 4, dtype=bf, device=device)
         self.act1 = torch.empty(pop * E, 20 * 20 * 16, dtype=bf,
                                 device=device)
-        self.act2 = torch.empty(pop * E, 2592, dtype=bf, device=device)
+        self.act2 = torch.empty(pop * E, 2592, dtype=torch.uint8,
+                                device=device)
         self.act3 = torch.empty(pop * E, 256, dtype=bf, device=device)
         self.state = torch.empty(pop * E, 4, dtype=torch.float32,
                                  device=device)
@@ -134,16 +139,17 @@ class ConvESEngine:
         iterp = self._iter_buf.data_ptr()
         o.es_perturb(self.theta.data_ptr(), NP_CONV, o.NP_CONV_PAD,
                      cfg.sigma, cfg.seed, iterp, member_offset, pop,
-                     self.wpert.data_ptr(), stream)
+                     self.wpert.data_ptr(), self.w3_fp8.data_ptr(), stream)
         o.conv_env_init(cfg.seed, iterp, pop, self.state.data_ptr(),
                         self.racc.data_ptr(), stream)
         for t in range(cfg.horizon):
             o.conv_obsgen(self.state.data_ptr(), self.gtab.data_ptr(),
                           cfg.seed, iterp, t, pop * cfg.envs_per_member,
                           self.obs.data_ptr(), stream)
-            o.conv_forward(self.wpert.data_ptr(), self.obs.data_ptr(),
-                           self.act1.data_ptr(), self.act2.data_ptr(),
-                           self.act3.data_ptr(), pop, stream)
+            o.conv_forward(self.wpert.data_ptr(), self.w3_fp8.data_ptr(),
+                           self.obs.data_ptr(), self.act1.data_ptr(),
+                           self.act2.data_ptr(), self.act3.data_ptr(),
+                           pop, stream)
             o.conv_head_env(self.wpert.data_ptr(), self.act3.data_ptr(),
                             pop, self.env_A.data_ptr(),
                             self.env_B.data_ptr(), self.state.data_ptr(),
@@ -277,27 +283,35 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
         philox_ref.env_init_state(seed, iteration, E)
     )
 
+    def fp8r(t):
+        return t.to(torch.float8_e4m3fn).to(torch.float32)
+
     for m in members:
         pair = m // 2
         sgn = -sigma if (m % 2) else sigma
         eps = torch.from_numpy(
             philox_ref.noise_for_pair(seed, iteration, pair, NP_CONV)
         )
-        th = bf(theta + sgn * eps)  # wpert materializes in bf16
+        th_raw = theta + sgn * eps
+        th = bf(th_raw)  # wpert materializes in bf16
 
-        def get(name):
+        def get(name, src=None):
             off, shape = OFF[name]
             n = 1
             for s in shape:
                 n *= s
-            return th[off : off + n].view(*shape)
+            return (src if src is not None else th)[off : off + n].view(
+                *shape
+            )
 
         # weight k-orders -> torch conv layout [oc][ic][ky][kx]
         w1 = get("w1").permute(0, 3, 1, 2).contiguous()
         b1 = get("b1")
         w2 = get("w2").permute(0, 3, 1, 2).contiguous()
         b2 = get("b2")
-        w3, b3 = get("w3"), get("b3")
+        # fc weights quantize to OCP e4m3 straight from fp32 (es_perturb)
+        w3 = fp8r(get("w3", th_raw))
+        b3 = get("b3")
         w4, b4 = get("w4"), get("b4")
 
         s = s0.clone()
@@ -309,9 +323,13 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
             x = obs.reshape(E, 84, 84, 4).permute(0, 3, 1, 2)
             h1 = bf(torch.tanh(
                 torch.nn.functional.conv2d(x, w1, b1, stride=4)))
-            h2 = bf(torch.tanh(
-                torch.nn.functional.conv2d(h1, w2, b2, stride=2)))
-            flat = h2.permute(0, 2, 3, 1).reshape(E, -1)  # (y*9+x)*32+oc
+            # layer-2 output goes straight from fp32 to fp8 (the kernel's
+            # epilogue does a single e4m3 rounding, no bf16 step)
+            h2 = torch.tanh(
+                torch.nn.functional.conv2d(h1, w2, b2, stride=2))
+            flat = fp8r(
+                h2.permute(0, 2, 3, 1).reshape(E, -1)
+            )  # (y*9+x)*32+oc, fp8 activations
             h3 = bf(torch.tanh(flat @ w3.T + b3))
             logits = h3 @ w4.T + b4
             action = logits.argmax(dim=1).float()
