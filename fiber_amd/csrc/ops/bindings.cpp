@@ -167,7 +167,7 @@ static void launch_conv_forward(uintptr_t wpert, uintptr_t w3_fp8,
                      (const __hip_bfloat16*)act1, nenv,
                      (unsigned char*)act2);
   check(hipGetLastError(), "conv_layer2 launch");
-  hipLaunchKernelGGL(conv_fc, dim3(nmembers), dim3(256), 0,
+  hipLaunchKernelGGL(conv_fc, dim3(nmembers, 4), dim3(256), 0,
                      (hipStream_t)stream, (const __hip_bfloat16*)wpert,
                      (const unsigned char*)w3_fp8,
                      (const unsigned char*)act2, nmembers,

@@ -305,30 +305,47 @@ conv_fc(const __hip_bfloat16* __restrict__ wpert,
   const int kgrp = lane >> 4;
   const int env = lane & 15;  // B column
 
-  for (int mt = wave; mt < 16; mt += 4) {
-    const int arow = mt * 16 + (lane & 15);
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int kk = 0; kk < 81; ++kk) {
-      const fp8x8 afrag = *reinterpret_cast<const fp8x8*>(
-          &w3[(size_t)arow * NFLAT + kk * 32 + kgrp * 8]);
-      const fp8x8 bfrag = *reinterpret_cast<const fp8x8*>(
-          &in[(size_t)env * NFLAT + kk * 32 + kgrp * 8]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag, bfrag, acc,
-                                                       0, 0, 0);
-    }
-    const int drow = mt * 16 + kgrp * 4;
-    union {
-      __hip_bfloat16 h[4];
-      unsigned long long u;
-    } pk;
-#pragma unroll
-    for (int ri = 0; ri < 4; ++ri) {
-      const float bias = __bfloat162float(wm[COFF_B3 + drow + ri]);
-      pk.h[ri] = __float2bfloat16(fast_tanh_c(acc[ri] + bias));
-    }
-    // D col = env, rows drow..drow+3 -> act3[env][drow..]
-    *reinterpret_cast<unsigned long long*>(&out[env * FCU + drow]) = pk.u;
+  // grid = (members, 4): each workgroup owns 4 of the 16 M-tiles, one
+  // per wave — 4x the workgroups of the one-wg-per-member mapping.
+  // Profiling showed this kernel latency-bound on the small 8 B
+  // fragment loads at 2 wg/CU, not HBM-BW-bound; more resident waves +
+  // 4-deep load batching is the fix.
+  const int mt = blockIdx.y * 4 + wave;
+  const int arow = mt * 16 + (lane & 15);
+  const unsigned char* ap = &w3[(size_t)arow * NFLAT + kgrp * 8];
+  const unsigned char* bp = &in[(size_t)env * NFLAT + kgrp * 8];
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int kk = 0; kk < 80; kk += 4) {
+    const fp8x8 a0 = *reinterpret_cast<const fp8x8*>(ap + (kk + 0) * 32);
+    const fp8x8 b0 = *reinterpret_cast<const fp8x8*>(bp + (kk + 0) * 32);
+    const fp8x8 a1 = *reinterpret_cast<const fp8x8*>(ap + (kk + 1) * 32);
+    const fp8x8 b1 = *reinterpret_cast<const fp8x8*>(bp + (kk + 1) * 32);
+    const fp8x8 a2 = *reinterpret_cast<const fp8x8*>(ap + (kk + 2) * 32);
+    const fp8x8 b2 = *reinterpret_cast<const fp8x8*>(bp + (kk + 2) * 32);
+    const fp8x8 a3 = *reinterpret_cast<const fp8x8*>(ap + (kk + 3) * 32);
+    const fp8x8 b3 = *reinterpret_cast<const fp8x8*>(bp + (kk + 3) * 32);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b2, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b3, acc, 0, 0, 0);
   }
+  {  // tail (81st K-tile)
+    const fp8x8 a = *reinterpret_cast<const fp8x8*>(ap + 80 * 32);
+    const fp8x8 b = *reinterpret_cast<const fp8x8*>(bp + 80 * 32);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc, 0, 0, 0);
+  }
+  const int drow = mt * 16 + kgrp * 4;
+  union {
+    __hip_bfloat16 h[4];
+    unsigned long long u;
+  } pk;
+#pragma unroll
+  for (int ri = 0; ri < 4; ++ri) {
+    const float bias = __bfloat162float(wm[COFF_B3 + drow + ri]);
+    pk.h[ri] = __float2bfloat16(fast_tanh_c(acc[ri] + bias));
+  }
+  // D col = env, rows drow..drow+3 -> act3[env][drow..]
+  *reinterpret_cast<unsigned long long*>(&out[env * FCU + drow]) = pk.u;
 }
 
 // ---------------------------------------------------------------------------
