@@ -30,14 +30,16 @@ extern "C" __global__ void mfma_gemm64_probe(const float*, const float*,
 // conv policy pipeline (conv_kernels.hip)
 extern "C" __global__ void es_perturb(const float*, int, int, float,
                                       uint32_t, const uint32_t*, int,
-                                      __hip_bfloat16*, unsigned char*);
+                                      __hip_bfloat16*, unsigned char*,
+                                      unsigned char*);
 extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
                                          float*, float*);
 extern "C" __global__ void conv_obsgen(const float*, const float*, uint32_t,
                                        const uint32_t*, uint32_t,
-                                       __hip_bfloat16*);
+                                       unsigned char*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
-                                       const __hip_bfloat16*, int,
+                                       const unsigned char*,
+                                       const unsigned char*, int,
                                        __hip_bfloat16*);
 extern "C" __global__ void conv_layer2(const __hip_bfloat16*,
                                        const __hip_bfloat16*, int,
@@ -121,13 +123,14 @@ static void launch_gemm64_probe(uintptr_t a, uintptr_t b, uintptr_t c,
 static void launch_perturb(uintptr_t theta, int nparams, int np_pad,
                            double sigma, uint32_t seed, uintptr_t iterp,
                            int member_offset, int pop, uintptr_t wpert,
-                           uintptr_t w3_fp8, uintptr_t stream) {
+                           uintptr_t w3_fp8, uintptr_t w1_fp8,
+                           uintptr_t stream) {
   const int bx = 64;  // grid-stride over param blocks
   hipLaunchKernelGGL(es_perturb, dim3(bx, pop), dim3(256), 0,
                      (hipStream_t)stream, (const float*)theta, nparams,
                      np_pad, (float)sigma, seed, (const uint32_t*)iterp,
                      member_offset, (__hip_bfloat16*)wpert,
-                     (unsigned char*)w3_fp8);
+                     (unsigned char*)w3_fp8, (unsigned char*)w1_fp8);
   check(hipGetLastError(), "es_perturb launch");
 }
 
@@ -148,18 +151,20 @@ static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
   hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
                      (hipStream_t)stream, (const float*)state,
                      (const float*)gtab, seed, (const uint32_t*)iterp, t,
-                     (__hip_bfloat16*)obs);
+                     (unsigned char*)obs);
   check(hipGetLastError(), "conv_obsgen launch");
 }
 
 static void launch_conv_forward(uintptr_t wpert, uintptr_t w3_fp8,
-                                uintptr_t obs, uintptr_t act1,
-                                uintptr_t act2, uintptr_t act3,
-                                int nmembers, uintptr_t stream) {
+                                uintptr_t w1_fp8, uintptr_t obs,
+                                uintptr_t act1, uintptr_t act2,
+                                uintptr_t act3, int nmembers,
+                                uintptr_t stream) {
   const int nenv = nmembers * 16;
   hipLaunchKernelGGL(conv_layer1, dim3(nenv), dim3(256), 0,
                      (hipStream_t)stream, (const __hip_bfloat16*)wpert,
-                     (const __hip_bfloat16*)obs, nenv,
+                     (const unsigned char*)w1_fp8,
+                     (const unsigned char*)obs, nenv,
                      (__hip_bfloat16*)act1);
   check(hipGetLastError(), "conv_layer1 launch");
   hipLaunchKernelGGL(conv_layer2, dim3(nenv), dim3(256), 0,
@@ -214,7 +219,8 @@ PYBIND11_MODULE(_ops, m) {
   m.def("es_perturb", &launch_perturb, py::arg("theta"), py::arg("nparams"),
         py::arg("np_pad"), py::arg("sigma"), py::arg("seed"),
         py::arg("iterp"), py::arg("member_offset"), py::arg("pop"),
-        py::arg("wpert"), py::arg("w3_fp8"), py::arg("stream"));
+        py::arg("wpert"), py::arg("w3_fp8"), py::arg("w1_fp8"),
+        py::arg("stream"));
   m.def("conv_env_init", &launch_conv_env_init, py::arg("seed"),
         py::arg("iterp"), py::arg("nmembers"), py::arg("state"),
         py::arg("racc"), py::arg("stream"));
@@ -222,9 +228,9 @@ PYBIND11_MODULE(_ops, m) {
         py::arg("gtab"), py::arg("seed"), py::arg("iterp"), py::arg("t"),
         py::arg("nenv_total"), py::arg("obs"), py::arg("stream"));
   m.def("conv_forward", &launch_conv_forward, py::arg("wpert"),
-        py::arg("w3_fp8"), py::arg("obs"), py::arg("act1"),
-        py::arg("act2"), py::arg("act3"), py::arg("nmembers"),
-        py::arg("stream"));
+        py::arg("w3_fp8"), py::arg("w1_fp8"), py::arg("obs"),
+        py::arg("act1"), py::arg("act2"), py::arg("act3"),
+        py::arg("nmembers"), py::arg("stream"));
   m.def("conv_head_env", &launch_conv_head_env, py::arg("wpert"),
         py::arg("act3"), py::arg("nmembers"), py::arg("env_A"),
         py::arg("env_B"), py::arg("state"), py::arg("racc"),

@@ -76,7 +76,8 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
                                       const uint32_t* __restrict__ iterp,
                                       int member_offset,
                                       __hip_bfloat16* __restrict__ wpert,
-                                      unsigned char* __restrict__ w3_fp8) {
+                                      unsigned char* __restrict__ w3_fp8,
+                                      unsigned char* __restrict__ w1_fp8) {
   const uint32_t iter = *iterp;
   const int member = member_offset + blockIdx.y;
   const uint32_t pair = (uint32_t)(member >> 1);
@@ -84,6 +85,8 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
   __hip_bfloat16* out = wpert + (size_t)blockIdx.y * np_pad;
   unsigned char* out8 =
       w3_fp8 ? w3_fp8 + (size_t)blockIdx.y * (FCU * NFLAT) : nullptr;
+  unsigned char* out8w1 =
+      w1_fp8 ? w1_fp8 + (size_t)blockIdx.y * (C1 * 256) : nullptr;
   const int jb0 = blockIdx.x * blockDim.x + threadIdx.x;
   const int stride = gridDim.x * blockDim.x;
   for (int jb = jb0; jb * 4 < nparams; jb += stride) {
@@ -101,6 +104,8 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
         // BW-bound on single-use weights)
         if (out8 && j >= COFF_W3 && j < COFF_B3)
           out8[j - COFF_W3] = __hip_fp8_e4m3(v).__x;
+        if (out8w1 && j < COFF_B1)
+          out8w1[j] = __hip_fp8_e4m3(v).__x;
       }
     }
   }
@@ -135,11 +140,11 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
                                        uint32_t seed,
                                        const uint32_t* __restrict__ iterp,
                                        uint32_t t,
-                                       __hip_bfloat16* __restrict__ obs) {
+                                       unsigned char* __restrict__ obs) {
   const uint32_t iter = *iterp;
   const int be = blockIdx.x;  // member*CENV + env
   const int e = be % CENV;
-  __hip_bfloat16* out = obs + (size_t)be * (IMG * IMG * CIN);
+  unsigned char* out = obs + (size_t)be * (IMG * IMG * CIN);
   float s[SDIM];
 #pragma unroll
   for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
@@ -149,13 +154,13 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
     fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)p, FAM_TAG_OBS, t, z);
     const float g = gtab[p];
     union {
-      __hip_bfloat16 h[4];
-      unsigned long long u;
+      unsigned char b[4];
+      uint32_t u;
     } pk;
 #pragma unroll
     for (int c = 0; c < CIN; ++c)
-      pk.h[c] = __float2bfloat16(0.52f * z[c] + s[c] * g);
-    *reinterpret_cast<unsigned long long*>(&out[p * CIN]) = pk.u;
+      pk.b[c] = __hip_fp8_e4m3(0.52f * z[c] + s[c] * g).__x;
+    *reinterpret_cast<uint32_t*>(&out[p * CIN]) = pk.u;
   }
 }
 
@@ -165,20 +170,22 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(256)
 conv_layer1(const __hip_bfloat16* __restrict__ wpert,
-            const __hip_bfloat16* __restrict__ obs, int nenv_total,
+            const unsigned char* __restrict__ w1_fp8,
+            const unsigned char* __restrict__ obs, int nenv_total,
             __hip_bfloat16* __restrict__ act1) {
-  __shared__ alignas(16) __hip_bfloat16 w1[C1][256];
+  __shared__ alignas(16) unsigned char w1[C1][256];
   __shared__ float b1[C1];
   const int be = blockIdx.x;
   const int member = be / CENV;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
-  const __hip_bfloat16* ob = obs + (size_t)be * (IMG * IMG * CIN);
+  const unsigned char* ob = obs + (size_t)be * (IMG * IMG * CIN);
   __hip_bfloat16* out = act1 + (size_t)be * (O1 * O1 * C1);
 
   const int tid = threadIdx.x;
   for (int i = tid; i < C1 * 256 / 8; i += blockDim.x) {
-    reinterpret_cast<bf16x8*>(&w1[0][0])[i] =
-        reinterpret_cast<const bf16x8*>(wm + COFF_W1)[i];
+    reinterpret_cast<fp8x8*>(&w1[0][0])[i] =
+        reinterpret_cast<const fp8x8*>(w1_fp8 +
+                                       (size_t)member * (C1 * 256))[i];
   }
   if (tid < C1) b1[tid] = __bfloat162float(wm[COFF_B1 + tid]);
   __syncthreads();
@@ -188,22 +195,33 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
   const int kgrp = lane >> 4;
   const int arow = lane & 15;  // output channel
 
-  bf16x8 afrag[8];
+  fp8x8 afrag[8];
 #pragma unroll
   for (int kk = 0; kk < 8; ++kk)
     afrag[kk] =
-        *reinterpret_cast<const bf16x8*>(&w1[arow][kk * 32 + kgrp * 8]);
+        *reinterpret_cast<const fp8x8*>(&w1[arow][kk * 32 + kgrp * 8]);
 
-  for (int nt = wave; nt < 25; nt += 4) {
-    const int pos = nt * 16 + (lane & 15);
-    const int oy = pos / O1, ox = pos % O1;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  // two N-tiles in flight per wave so the scattered 8 B obs loads of one
+  // tile hide under the other's MFMAs (latency-bound otherwise)
+  const int c = lane & 15;
+  for (int nt = wave; nt < 25; nt += 8) {
+    const int nt2 = nt + 4;
+    const bool two = nt2 < 25;
+    const int pos0 = nt * 16 + c, oy0 = pos0 / O1, ox0 = pos0 % O1;
+    const int p1 = two ? nt2 * 16 + c : pos0;
+    const int oy1 = p1 / O1, ox1 = p1 % O1;
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {  // ky = kk
-      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-          &ob[(((oy * 4 + kk) * IMG) + ox * 4 + kgrp * 2) * CIN]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bfrag, acc,
-                                                    0, 0, 0);
+      const fp8x8 b0 = *reinterpret_cast<const fp8x8*>(
+          &ob[(((oy0 * 4 + kk) * IMG) + ox0 * 4 + kgrp * 2) * CIN]);
+      const fp8x8 b1f = *reinterpret_cast<const fp8x8*>(
+          &ob[(((oy1 * 4 + kk) * IMG) + ox1 * 4 + kgrp * 2) * CIN]);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag[kk], b0,
+                                                        acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag[kk], b1f,
+                                                        acc1, 0, 0, 0);
     }
     const int drow = kgrp * 4;  // output-channel base for this lane's D
     union {
@@ -212,8 +230,15 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
     } pk;
 #pragma unroll
     for (int ri = 0; ri < 4; ++ri)
-      pk.h[ri] = __float2bfloat16(fast_tanh_c(acc[ri] + b1[drow + ri]));
-    *reinterpret_cast<unsigned long long*>(&out[pos * C1 + drow]) = pk.u;
+      pk.h[ri] = __float2bfloat16(fast_tanh_c(acc0[ri] + b1[drow + ri]));
+    *reinterpret_cast<unsigned long long*>(&out[pos0 * C1 + drow]) = pk.u;
+    if (two) {
+#pragma unroll
+      for (int ri = 0; ri < 4; ++ri)
+        pk.h[ri] =
+            __float2bfloat16(fast_tanh_c(acc1[ri] + b1[drow + ri]));
+      *reinterpret_cast<unsigned long long*>(&out[p1 * C1 + drow]) = pk.u;
+    }
   }
 }
 

@@ -110,7 +110,12 @@ class ConvESEngine:
         # HBM-bound on single-use weights, fp8 halves the traffic
         self.w3_fp8 = torch.empty(pop, 256 * 2592, dtype=torch.uint8,
                                   device=device)
-        self.obs = torch.empty(pop * E, 84 * 84 * 4, dtype=bf, device=device)
+        self.w1_fp8 = torch.empty(pop, 16 * 256, dtype=torch.uint8,
+                                  device=device)
+        # pixel observations also ride fp8 (obsgen/conv1 were ~45% of the
+        # step and BW/latency-bound on the 84x84x4 obs buffers)
+        self.obs = torch.empty(pop * E, 84 * 84 * 4, dtype=torch.uint8,
+                               device=device)
         self.act1 = torch.empty(pop * E, 20 * 20 * 16, dtype=bf,
                                 device=device)
         self.act2 = torch.empty(pop * E, 2592, dtype=torch.uint8,
@@ -139,7 +144,8 @@ class ConvESEngine:
         iterp = self._iter_buf.data_ptr()
         o.es_perturb(self.theta.data_ptr(), NP_CONV, o.NP_CONV_PAD,
                      cfg.sigma, cfg.seed, iterp, member_offset, pop,
-                     self.wpert.data_ptr(), self.w3_fp8.data_ptr(), stream)
+                     self.wpert.data_ptr(), self.w3_fp8.data_ptr(),
+                     self.w1_fp8.data_ptr(), stream)
         o.conv_env_init(cfg.seed, iterp, pop, self.state.data_ptr(),
                         self.racc.data_ptr(), stream)
         for t in range(cfg.horizon):
@@ -147,9 +153,9 @@ class ConvESEngine:
                           cfg.seed, iterp, t, pop * cfg.envs_per_member,
                           self.obs.data_ptr(), stream)
             o.conv_forward(self.wpert.data_ptr(), self.w3_fp8.data_ptr(),
-                           self.obs.data_ptr(), self.act1.data_ptr(),
-                           self.act2.data_ptr(), self.act3.data_ptr(),
-                           pop, stream)
+                           self.w1_fp8.data_ptr(), self.obs.data_ptr(),
+                           self.act1.data_ptr(), self.act2.data_ptr(),
+                           self.act3.data_ptr(), pop, stream)
             o.conv_head_env(self.wpert.data_ptr(), self.act3.data_ptr(),
                             pop, self.env_A.data_ptr(),
                             self.env_B.data_ptr(), self.state.data_ptr(),
@@ -305,7 +311,8 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
             )
 
         # weight k-orders -> torch conv layout [oc][ic][ky][kx]
-        w1 = get("w1").permute(0, 3, 1, 2).contiguous()
+        # conv1 weights quantize to e4m3 straight from fp32 (es_perturb)
+        w1 = fp8r(get("w1", th_raw)).permute(0, 3, 1, 2).contiguous()
         b1 = get("b1")
         w2 = get("w2").permute(0, 3, 1, 2).contiguous()
         b2 = get("b2")
@@ -319,7 +326,7 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
         for t in range(horizon):
             z = obs_noise(t)  # [E][7056][4]
             obs = 0.52 * z + s[:, None, :] * gtab.reshape(1, -1, 1)
-            obs = bf(obs)
+            obs = fp8r(obs)
             x = obs.reshape(E, 84, 84, 4).permute(0, 3, 1, 2)
             h1 = bf(torch.tanh(
                 torch.nn.functional.conv2d(x, w1, b1, stride=4)))
