@@ -33,9 +33,11 @@ class TestConvPipeline:
         iterp = torch.zeros(1, dtype=torch.int32, device=device)
         w3_fp8 = torch.empty(pop, 256 * 2592, dtype=torch.uint8,
                              device=device)
+        w1_fp8 = torch.empty(pop, 16 * 256, dtype=torch.uint8,
+                             device=device)
         o.es_perturb(theta.data_ptr(), conv_policy.NP_CONV, o.NP_CONV_PAD,
                      0.02, 11, iterp.data_ptr(), 0, pop, wpert.data_ptr(),
-                     w3_fp8.data_ptr(),
+                     w3_fp8.data_ptr(), w1_fp8.data_ptr(),
                      torch.cuda.current_stream().cuda_stream)
         torch.cuda.synchronize()
         for m in (0, 1, 3):
