@@ -3,6 +3,14 @@ N messages through SimpleQueue across two processes, msgs/sec + MB/s,
 fiber_amd shm rings vs stdlib multiprocessing.
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import argparse
 import multiprocessing
 import time

@@ -10,6 +10,14 @@ multiprocessing.Pool on the same machine.
 Run: python examples/dispatch_bench.py [--workers 8]
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import argparse
 import multiprocessing
 import time

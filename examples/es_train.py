@@ -5,6 +5,14 @@ Multi GPU:   python -m torch.distributed.run --nproc-per-node 8 \
                  --master-addr 127.0.0.1 examples/es_train.py --iters 100
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import argparse
 import os
 import time

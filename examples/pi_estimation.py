@@ -4,6 +4,14 @@ Reference analog: uber/fiber examples/pi_estimation.py (Pool(4).map over
 1e6 samples on the local backend, CPU plumbing only).
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import random
 import time
 

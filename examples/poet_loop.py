@@ -10,6 +10,14 @@ Runs on CPU plumbing (the per-pair payload here is a small numpy ES
 step) or with @fiber_amd.meta(gpu=1) workers when GPUs are present.
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import argparse
 
 import numpy as np

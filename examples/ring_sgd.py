@@ -6,6 +6,14 @@ CPU demo: python examples/ring_sgd.py            (gloo, world 2)
 GPU:      python examples/ring_sgd.py --gpus 4   (RCCL over xGMI)
 """
 
+import os as _os
+import sys as _sys
+
+_REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+if _REPO_ROOT not in _sys.path:
+    _sys.path.insert(0, _REPO_ROOT)
+
+
 import argparse
 import functools
 
