@@ -35,7 +35,10 @@ def main():
                              "ConvNet pixel-policy config")
     args = parser.parse_args()
     if args.pop_per_gpu is None:
-        args.pop_per_gpu = 4096 if args.model == "mlp" else 512
+        # Big shards are the MI355X-first choice (288 GB HBM, fixed costs
+        # amortize, collectives stay large); per-GPU work is fixed as N
+        # grows (weak scaling).
+        args.pop_per_gpu = 8192 if args.model == "mlp" else 512
     if args.horizon is None:
         args.horizon = 256 if args.model == "mlp" else 64
 

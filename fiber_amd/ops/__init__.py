@@ -85,12 +85,19 @@ def es_grad(wpair, pair_begin, pair_end, seed, iteration, device,
 
 
 def centered_rank(fitness):
-    """Centered rank transform in [-0.5, 0.5]."""
-    ops = _require_ops()
+    """Centered rank transform in [-0.5, 0.5].
+
+    The O(n^2) comparison kernel wins for small populations (one launch,
+    no sort); past ~16k members the argsort path (device radix sort) is
+    asymptotically required — at 64k members the n^2 kernel would cost
+    milliseconds."""
     _check(fitness, "fitness")
+    n = fitness.numel()
+    if n > 16384:
+        return centered_rank_ref(fitness)
+    ops = _require_ops()
     out = torch.empty_like(fitness)
-    ops.centered_rank(fitness.data_ptr(), fitness.numel(), out.data_ptr(),
-                      _stream())
+    ops.centered_rank(fitness.data_ptr(), n, out.data_ptr(), _stream())
     return out
 
 
