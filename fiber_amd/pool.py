@@ -206,7 +206,12 @@ class AsyncResult:
     def successful(self):
         if not self.ready():
             raise ValueError("result not ready")
-        return True
+        job = self._pool._inventory._jobs.get(self._seq)
+        if job is None:
+            return True  # already collected without error
+        return job["error"] is None and not any(
+            isinstance(r, _ExcInfo) for r in job["results"]
+        )
 
 
 MapResult = AsyncResult

@@ -108,6 +108,23 @@ class TestProcess:
         p.join(30)
 
 
+class TestManyFds:
+    def test_spawn_with_1100_open_fds(self):
+        """Reference regression (tests/test_popen.py:95-113): process
+        management must not depend on select()-able fd numbers < 1024."""
+        import os
+
+        fds = [os.open("/dev/null", os.O_RDONLY) for _ in range(1100)]
+        try:
+            p = fiber_amd.Process(target=_noop)
+            p.start()
+            p.join(60)
+            assert p.exitcode == 0
+        finally:
+            for fd in fds:
+                os.close(fd)
+
+
 class TestStartFailure:
     def test_backend_start_timeout_surfaces(self, monkeypatch):
         """Fault injection at the backend seam (reference TimeoutBackend
