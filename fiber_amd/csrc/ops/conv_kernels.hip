@@ -148,19 +148,24 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
   float s[SDIM];
 #pragma unroll
   for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
-  // each position needs CIN=4 values; one philox draw per position
-  for (int p = threadIdx.x; p < IMG * IMG; p += blockDim.x) {
-    float z[4];
-    fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)p, FAM_TAG_OBS, t, z);
-    const float g = gtab[p];
+  // each position needs CIN=4 values (one philox draw); two adjacent
+  // positions per thread iteration -> one 8 B store
+  for (int p = threadIdx.x * 2; p < IMG * IMG; p += blockDim.x * 2) {
     union {
-      unsigned char b[4];
-      uint32_t u;
+      unsigned char b[8];
+      unsigned long long u;
     } pk;
 #pragma unroll
-    for (int c = 0; c < CIN; ++c)
-      pk.b[c] = __hip_fp8_e4m3(0.52f * z[c] + s[c] * g).__x;
-    *reinterpret_cast<uint32_t*>(&out[p * CIN]) = pk.u;
+    for (int q = 0; q < 2; ++q) {
+      float z[4];
+      fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)(p + q), FAM_TAG_OBS,
+                   t, z);
+      const float g = gtab[p + q];
+#pragma unroll
+      for (int c = 0; c < CIN; ++c)
+        pk.b[q * 4 + c] = __hip_fp8_e4m3(0.52f * z[c] + s[c] * g).__x;
+    }
+    *reinterpret_cast<unsigned long long*>(&out[p * CIN]) = pk.u;
   }
 }
 
