@@ -30,11 +30,12 @@ __device__ __host__ inline uint32_t fam_mulhi(uint32_t a, uint32_t b) {
 #endif
 }
 
-__device__ __host__ inline fam_uint4 philox4x32_10(uint32_t k0, uint32_t k1,
-                                                   uint32_t c0, uint32_t c1,
-                                                   uint32_t c2, uint32_t c3) {
+template <int ROUNDS>
+__device__ __host__ inline fam_uint4 philox4x32_r(uint32_t k0, uint32_t k1,
+                                                  uint32_t c0, uint32_t c1,
+                                                  uint32_t c2, uint32_t c3) {
 #pragma unroll
-  for (int r = 0; r < 10; ++r) {
+  for (int r = 0; r < ROUNDS; ++r) {
     uint32_t hi0 = fam_mulhi(FAM_PHILOX_M0, c0);
     uint32_t lo0 = FAM_PHILOX_M0 * c0;
     uint32_t hi1 = fam_mulhi(FAM_PHILOX_M1, c2);
@@ -53,14 +54,24 @@ __device__ __host__ inline fam_uint4 philox4x32_10(uint32_t k0, uint32_t k1,
   return {c0, c1, c2, c3};
 }
 
+__device__ __host__ inline fam_uint4 philox4x32_10(uint32_t k0, uint32_t k1,
+                                                   uint32_t c0, uint32_t c1,
+                                                   uint32_t c2, uint32_t c3) {
+  return philox4x32_r<10>(k0, k1, c0, c1, c2, c3);
+}
+
 // 4 uniforms in [-1, 1) from one philox draw (no transcendentals —
 // used for the pixel-observation noise where gaussianity is not needed
 // and Box-Muller's log/sqrt/sincos dominated the obsgen kernel).
+// 7 rounds: the minimum Philox4x32 configuration that passes BigCrush
+// (Salmon et al. 2011, Table 2) — used for the bulk pixel-obs noise
+// where the 32-bit mulhi chains dominate the obsgen kernel's VALU time.
+// ES perturbations keep the default 10 rounds.
 __device__ __host__ inline void fam_uniform4(uint32_t k0, uint32_t k1,
                                              uint32_t c0, uint32_t c1,
                                              uint32_t c2, uint32_t c3,
                                              float z[4]) {
-  fam_uint4 u = philox4x32_10(k0, k1, c0, c1, c2, c3);
+  fam_uint4 u = philox4x32_r<7>(k0, k1, c0, c1, c2, c3);
   const float inv31 = 4.656612873077393e-10f;  // 2^-31
   z[0] = (float)(int32_t)u.x * inv31;
   z[1] = (float)(int32_t)u.y * inv31;

@@ -130,36 +130,60 @@ class ConvESEngine:
         self._iter_buf = torch.zeros(1, dtype=torch.int32, device=device)
         self._graph = None
         self.use_graph = True
+        nhalves = 2 if pop % 2 == 0 and pop >= 8 else 1
+        self._half_streams = [
+            torch.cuda.Stream(device=device) for _ in range(nhalves)
+        ]
 
     def _stream(self):
         return torch.cuda.current_stream().cuda_stream
 
-    def _rollout_body(self):
-        """The full rollout kernel DAG (captured into a hipGraph)."""
+    def _rollout_half(self, half, nhalves):
+        """One population slice's full rollout chain (kernels on the
+        caller's current stream; slices are disjoint, so two halves on
+        two streams are fully independent — the scheduler overlaps the
+        VALU-bound obsgen with the BW-bound fc/conv phases of the other
+        half)."""
         o = ops._require_ops()
         cfg = self.cfg
-        pop = cfg.pop_per_gpu
-        member_offset = self.rank * pop
+        E = cfg.envs_per_member
+        pop = cfg.pop_per_gpu // nhalves
+        m0 = half * pop  # local member base
+        member_offset = self.rank * cfg.pop_per_gpu + m0
         stream = self._stream()
         iterp = self._iter_buf.data_ptr()
+        wpert = self.wpert[m0:].data_ptr()
+        w3_fp8 = self.w3_fp8[m0:].data_ptr()
+        w1_fp8 = self.w1_fp8[m0:].data_ptr()
+        obs = self.obs[m0 * E:].data_ptr()
+        act1 = self.act1[m0 * E:].data_ptr()
+        act2 = self.act2[m0 * E:].data_ptr()
+        act3 = self.act3[m0 * E:].data_ptr()
+        state = self.state[m0 * E:].data_ptr()
+        racc = self.racc[m0:].data_ptr()
         o.es_perturb(self.theta.data_ptr(), NP_CONV, o.NP_CONV_PAD,
                      cfg.sigma, cfg.seed, iterp, member_offset, pop,
-                     self.wpert.data_ptr(), self.w3_fp8.data_ptr(),
-                     self.w1_fp8.data_ptr(), stream)
-        o.conv_env_init(cfg.seed, iterp, pop, self.state.data_ptr(),
-                        self.racc.data_ptr(), stream)
+                     wpert, w3_fp8, w1_fp8, stream)
+        o.conv_env_init(cfg.seed, iterp, pop, state, racc, stream)
         for t in range(cfg.horizon):
-            o.conv_obsgen(self.state.data_ptr(), self.gtab.data_ptr(),
-                          cfg.seed, iterp, t, pop * cfg.envs_per_member,
-                          self.obs.data_ptr(), stream)
-            o.conv_forward(self.wpert.data_ptr(), self.w3_fp8.data_ptr(),
-                           self.w1_fp8.data_ptr(), self.obs.data_ptr(),
-                           self.act1.data_ptr(), self.act2.data_ptr(),
-                           self.act3.data_ptr(), pop, stream)
-            o.conv_head_env(self.wpert.data_ptr(), self.act3.data_ptr(),
-                            pop, self.env_A.data_ptr(),
-                            self.env_B.data_ptr(), self.state.data_ptr(),
-                            self.racc.data_ptr(), stream)
+            o.conv_obsgen(state, self.gtab.data_ptr(), cfg.seed, iterp, t,
+                          pop * E, obs, stream)
+            o.conv_forward(wpert, w3_fp8, w1_fp8, obs, act1, act2, act3,
+                           pop, stream)
+            o.conv_head_env(wpert, act3, pop, self.env_A.data_ptr(),
+                            self.env_B.data_ptr(), state, racc, stream)
+
+    def _rollout_body(self):
+        """The full rollout kernel DAG (captured into a hipGraph): two
+        half-population chains forked onto two streams."""
+        main = torch.cuda.current_stream()
+        for s in self._half_streams:
+            s.wait_stream(main)
+        for half, s in enumerate(self._half_streams):
+            with torch.cuda.stream(s):
+                self._rollout_half(half, len(self._half_streams))
+        for s in self._half_streams:
+            main.wait_stream(s)
 
     def _ensure_graph(self):
         if self._graph is not None:

@@ -13,7 +13,7 @@ TAG_NOISE = np.uint32(0x45530001)
 TAG_ENV = np.uint32(0x45530002)
 
 
-def philox4x32_10(k0, k1, c0, c1, c2, c3):
+def philox4x32_10(k0, k1, c0, c1, c2, c3, rounds=10):
     """Vectorized over arrays of counters.  All args uint32 arrays/scalars."""
     k0 = np.uint32(k0) * np.ones_like(np.asarray(c0, np.uint32))
     k1 = np.uint32(k1) * np.ones_like(np.asarray(c0, np.uint32))
@@ -22,7 +22,7 @@ def philox4x32_10(k0, k1, c0, c1, c2, c3):
     c2 = np.asarray(c2, np.uint32) * np.ones_like(c0)
     c3 = np.asarray(c3, np.uint32) * np.ones_like(c0)
     with np.errstate(over="ignore"):
-        for _ in range(10):
+        for _ in range(rounds):
             prod0 = c0.astype(np.uint64) * np.uint64(_M0)
             prod1 = c2.astype(np.uint64) * np.uint64(_M1)
             hi0 = (prod0 >> np.uint64(32)).astype(np.uint32)
@@ -59,8 +59,9 @@ def normal4(k0, k1, c0, c1, c2, c3):
 
 
 def uniform4(k0, k1, c0, c1, c2, c3):
-    """4 uniforms in [-1, 1) per counter (matches fam_uniform4)."""
-    x0, x1, x2, x3 = philox4x32_10(k0, k1, c0, c1, c2, c3)
+    """4 uniforms in [-1, 1) per counter (matches fam_uniform4:
+    Philox4x32-7, the minimum BigCrush-passing round count)."""
+    x0, x1, x2, x3 = philox4x32_10(k0, k1, c0, c1, c2, c3, rounds=7)
     inv31 = np.float32(4.656612873077393e-10)
     z = [
         x.astype(np.int32).astype(np.float32) * inv31
