@@ -195,7 +195,6 @@ def rollout_reference(theta, sigma, seed, iteration, horizon, members,
                       obs_mu, obs_nu, env_A, env_B):
     """Returns fitness[len(members)] computed on CPU in fp32 (+bf16
     rounding at the same points as the kernel)."""
-    import numpy as np
 
     from . import philox_ref
 

@@ -34,7 +34,6 @@ def _watchdog(sock):
 
 def spawn_main(sock_path, ident):
     from . import config as fam_config
-    from . import process as fam_process
     from .popen import recv_msg, send_msg
     from . import serialization
 
