@@ -38,7 +38,7 @@ NP_CONV = 677686
 
 @dataclasses.dataclass
 class ConvESConfig:
-    pop_per_gpu: int = 512        # must be even
+    pop_per_gpu: int = 1024       # must be even
     horizon: int = 64
     sigma: float = 0.02
     lr: float = 0.01
