@@ -64,6 +64,7 @@ def main():
         cfg = ESConfig(pop_per_gpu=args.pop_per_gpu, horizon=args.horizon)
         engine = ESEngine(cfg, ctx=ctx, device=device)
         model_name = "es-mlp-obs4-h64x64-act2"
+        dtype = "bf16"
     else:
         from fiber_amd.es.conv_policy import ConvESConfig, ConvESEngine
 
@@ -71,6 +72,7 @@ def main():
                            horizon=args.horizon)
         engine = ConvESEngine(cfg, ctx=ctx, device=device)
         model_name = "es-conv-84x84x4-dqn-act6"
+        dtype = "bf16+fp8(e4m3)"  # conv1/fc ride OCP fp8, rest bf16
 
     for i in range(args.warmup):
         engine.step(iteration=i)
@@ -109,7 +111,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": dtype,
             "data": "synthetic",
             "config": {
                 "model": model_name,
