@@ -23,6 +23,7 @@
 #include <atomic>
 #include <cerrno>
 #include <cstdint>
+#include <cstdio>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -73,7 +74,20 @@ inline size_t record_bytes(size_t len) {
 
 inline bool process_alive(uint32_t pid) {
   if (pid == 0) return false;
-  return kill((pid_t)pid, 0) == 0 || errno != ESRCH;
+  if (kill((pid_t)pid, 0) != 0 && errno == ESRCH) return false;
+  // kill(0) reports zombies as alive; a zombie writer will never commit,
+  // so check the /proc state byte (field after the comm parens).
+  char path[64];
+  snprintf(path, sizeof(path), "/proc/%u/stat", pid);
+  FILE* f = fopen(path, "r");
+  if (!f) return false;
+  char buf[512];
+  size_t n = fread(buf, 1, sizeof(buf) - 1, f);
+  fclose(f);
+  buf[n] = 0;
+  const char* p = strrchr(buf, ')');
+  if (p && p[1] == ' ' && (p[2] == 'Z' || p[2] == 'X')) return false;
+  return true;
 }
 
 inline void make_deadline(double timeout_s, struct timespec* ts) {

@@ -13,7 +13,7 @@ import dataclasses
 
 import torch
 
-from .. import ops
+from .. import ops, tracing
 from .engine import make_env_params
 
 
@@ -214,10 +214,11 @@ class ConvESEngine:
         horizon x 5 launch + host-loop overheads; falls back to eager
         launches if graph capture is unavailable."""
         self._iter_buf.fill_(int(iteration))
-        if self._ensure_graph():
-            self._graph.replay()
-        else:
-            self._rollout_body()
+        with tracing.range("es.conv_rollout"):
+            if self._ensure_graph():
+                self._graph.replay()
+            else:
+                self._rollout_body()
         return self.racc.mean(dim=1)
 
     def step(self, iteration=None):

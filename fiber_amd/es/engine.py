@@ -19,7 +19,7 @@ import dataclasses
 
 import torch
 
-from .. import ops
+from .. import ops, tracing
 
 
 @dataclasses.dataclass
@@ -96,14 +96,16 @@ class ESEngine:
         shard = cfg.pop_per_gpu
         member_offset = self.rank * shard
 
-        fitness, obs_stat = ops.es_rollout_mlp(
-            self.theta, cfg.sigma, cfg.seed, iteration, cfg.horizon,
-            member_offset, shard, self.obs_mu, self.obs_nu, self.env_A,
-            self.env_B,
-        )
+        with tracing.range("es.rollout"):
+            fitness, obs_stat = ops.es_rollout_mlp(
+                self.theta, cfg.sigma, cfg.seed, iteration, cfg.horizon,
+                member_offset, shard, self.obs_mu, self.obs_nu, self.env_A,
+                self.env_B,
+            )
 
         if self.ctx is not None:
-            self.ctx.all_gather_into(self._fitness_all, fitness)
+            with tracing.range("es.allgather_fitness"):
+                self.ctx.all_gather_into(self._fitness_all, fitness)
             fitness_all = self._fitness_all
         else:
             fitness_all = fitness
@@ -113,11 +115,13 @@ class ESEngine:
 
         pair_begin = member_offset // 2
         pair_end = (member_offset + shard) // 2
-        grad = ops.es_grad(wpair, pair_begin, pair_end, cfg.seed, iteration,
-                           self.device)
+        with tracing.range("es.grad"):
+            grad = ops.es_grad(wpair, pair_begin, pair_end, cfg.seed,
+                               iteration, self.device)
         if self.ctx is not None:
-            self.ctx.allreduce(grad)
-            self.ctx.allreduce(obs_stat)
+            with tracing.range("es.allreduce_grad"):
+                self.ctx.allreduce(grad)
+                self.ctx.allreduce(obs_stat)
         grad /= float(self.pop_total) * cfg.sigma
 
         # maximize fitness => ascend

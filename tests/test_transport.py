@@ -134,3 +134,49 @@ class TestSocket:
     def test_bad_mode_raises(self):
         with pytest.raises(ValueError):
             Socket("xyz", "addr")
+
+
+def _flood_producer(addr, payload_size, stop_after):
+    from fiber_amd.transport import Socket
+
+    sock = Socket("w", addr, bind=False)
+    blob = b"p" * payload_size
+    for _ in range(stop_after):
+        sock.send(blob, timeout=5.0)
+
+
+class TestDeadWriterReclaim:
+    def test_consumer_survives_producer_sigkill_mid_stream(self):
+        """A producer SIGKILLed while streaming large payloads must not
+        wedge the ring: the reserve/commit protocol + dead-writer reclaim
+        keeps other producers' messages flowing."""
+        import time
+
+        import fiber_amd
+        from fiber_amd.transport import Socket, new_address
+
+        addr = new_address("fam-chaos")
+        reader = Socket("r", addr, bind=True)
+        victim = fiber_amd.Process(
+            target=_flood_producer, args=(addr, 100_000, 1_000_000)
+        )
+        survivor = fiber_amd.Process(
+            target=_flood_producer, args=(addr, 100_000, 200)
+        )
+        victim.start()
+        survivor.start()
+        # let them stream, then SIGKILL the victim mid-flight
+        for _ in range(20):
+            assert reader.recv(timeout=10.0) is not None
+        victim.kill()
+        # drain: within a bounded time we must observe the survivor's
+        # full 200 messages despite the victim's corpse in the ring
+        got = 0
+        deadline = time.monotonic() + 30.0
+        while got < 100 and time.monotonic() < deadline:
+            if reader.recv(timeout=5.0) is not None:
+                got += 1
+        assert got >= 100, "ring wedged after producer SIGKILL"
+        victim.join(10)
+        survivor.join(30)
+        reader.close()
