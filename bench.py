@@ -38,7 +38,7 @@ def main():
         # Big shards are the MI355X-first choice (288 GB HBM, fixed costs
         # amortize, collectives stay large); per-GPU work is fixed as N
         # grows (weak scaling).
-        args.pop_per_gpu = 8192 if args.model == "mlp" else 1024
+        args.pop_per_gpu = 16384 if args.model == "mlp" else 1024
     if args.horizon is None:
         args.horizon = 256 if args.model == "mlp" else 64
 

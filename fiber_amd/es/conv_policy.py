@@ -130,7 +130,11 @@ class ConvESEngine:
         self._iter_buf = torch.zeros(1, dtype=torch.int32, device=device)
         self._graph = None
         self.use_graph = True
-        nhalves = 2 if pop % 2 == 0 and pop >= 8 else 1
+        import os as _os
+
+        nhalves = int(_os.environ.get("FAM_CONV_STREAMS", "2"))
+        if pop % nhalves or pop < 4 * nhalves:
+            nhalves = 1
         self._half_streams = [
             torch.cuda.Stream(device=device) for _ in range(nhalves)
         ]
