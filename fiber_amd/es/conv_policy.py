@@ -132,7 +132,7 @@ class ConvESEngine:
         self.use_graph = True
         import os as _os
 
-        nhalves = int(_os.environ.get("FAM_CONV_STREAMS", "2"))
+        nhalves = int(_os.environ.get("FAM_CONV_STREAMS", "4"))
         if pop % nhalves or pop < 4 * nhalves:
             nhalves = 1
         self._half_streams = [
