@@ -77,6 +77,7 @@ setup(
         "fiber_amd.backends",
         "fiber_amd.ops",
         "fiber_amd.es",
+        "fiber_amd.experimental",
     ],
     ext_modules=[
         Extension(

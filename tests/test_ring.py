@@ -101,3 +101,12 @@ class TestRing:
         except RuntimeError:
             raised = True
         assert raised
+
+
+class TestExperimentalAlias:
+    def test_reference_import_path(self):
+        from fiber_amd.experimental.ring import Ring as R2
+        from fiber_amd.experimental.ring import RingNode as N2
+
+        assert R2 is Ring
+        assert N2 is RingNode
