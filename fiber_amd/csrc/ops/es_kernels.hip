@@ -119,6 +119,61 @@ __device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
   }
 }
 
+// Layer-2 strip with the logits fused into the MFMA epilogue: the
+// activations stay in fp32 registers (no bf16 pack / LDS store — they
+// feed nothing but the logits), each lane accumulates its 4 rows'
+// contribution to both action logits, and a 2-stage shfl_xor reduce over
+// the 4 kgrp lane groups yields the wave's per-env partial, written to
+// lpart[wave][env].  Removes the old D1 phase, its barrier, and all h2
+// LDS traffic.  w3r = this lane's 8 w3 weights (2 actions x 4 rows).
+template <int K, int AS, int BS>
+__device__ inline void mfma_strip_logits(
+    const __hip_bfloat16* __restrict__ A,
+    const __hip_bfloat16* __restrict__ B, const float* __restrict__ bias,
+    const float* __restrict__ w3r, float (*lpart)[ENVS][2], int wave,
+    int lane) {
+  const int r0 = wave * 16;
+  const int arow = r0 + (lane & 15);
+  const int kgrp = lane >> 4;
+
+  bf16x8 afrag[K / 32];
+#pragma unroll
+  for (int kk = 0; kk < K / 32; ++kk) {
+    afrag[kk] = *reinterpret_cast<const bf16x8*>(
+        &A[arow * AS + kk * 32 + kgrp * 8]);
+  }
+
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int bcol = ct * 16 + (lane & 15);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < K / 32; ++kk) {
+      bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &B[bcol * BS + kk * 32 + kgrp * 8]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bfrag, acc,
+                                                    0, 0, 0);
+    }
+    const int drow = r0 + kgrp * 4;
+    float p0 = 0.f, p1 = 0.f;
+#pragma unroll
+    for (int ri = 0; ri < 4; ++ri) {
+      const float h = fast_tanh(acc[ri] + bias[drow + ri]);
+      p0 += w3r[ri] * h;
+      p1 += w3r[4 + ri] * h;
+    }
+    // reduce over the 4 kgrp groups (lanes l, l^16, l^32 share bcol)
+    p0 += __shfl_xor(p0, 16, 64);
+    p1 += __shfl_xor(p1, 16, 64);
+    p0 += __shfl_xor(p0, 32, 64);
+    p1 += __shfl_xor(p1, 32, 64);
+    if (kgrp == 0) {
+      lpart[wave][bcol][0] = p0;
+      lpart[wave][bcol][1] = p1;
+    }
+  }
+}
+
 // Decompose flat theta index j -> LDS slot write.
 struct PolicyLds {
   alignas(16) __hip_bfloat16 w1[HID][S1];
@@ -150,18 +205,13 @@ __device__ inline void store_param(PolicyLds* p, int j, float v) {
 // Shared epilogue/prologue state for one member rollout.
 struct RolloutLds {
   PolicyLds pol;
-  // xb (layer-1 B operand, live phase A->B) aliases h2 (layer-2 output,
-  // live phase C->D1): their live ranges never overlap, and the union
-  // shrinks the block to ~39 KB => 4 workgroups/CU instead of 3.
-  // (Garbage in xb's K-padding from old h2 values is harmless: the
-  // W1-side padding is zero and old h2 values are finite, never NaN.)
-  union {
-    alignas(16) __hip_bfloat16 xb[ENVS][S1];
-    alignas(16) __hip_bfloat16 h2[ENVS][S2];
-  };
+  // layer-2 output never touches LDS (fused-logits epilogue keeps it in
+  // registers), so only the layer-1 B operand lives here (~35 KB block
+  // => 4 workgroups/CU).
+  alignas(16) __hip_bfloat16 xb[ENVS][S1];
   alignas(16) __hip_bfloat16 h1[ENVS][S2];  // B-operand layer 2
   float S[2][ENVS][OBS];      // env state (double-buffered)
-  float lpart[OBS][ENVS][2];  // logits partials (D1 -> D2; epilogue reuse)
+  float lpart[4][ENVS][2];  // per-wave logits partials (C -> D2)
   float ostat[2 * OBS + 1];   // sum, sumsq, count
 };
 
@@ -230,6 +280,17 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
   for (int e = 0; e < OBS; ++e) eA_d[e] = env_A[dd * OBS + e];
   const float eB_d = env_B[dd];
   const float one_if_d0 = (dd == 0) ? 1.f : 0.f;
+  // this lane's w3 slice for the fused logits epilogue: rows
+  // drow..drow+3 of both actions (drow = wave*16 + kgrp*4)
+  float w3r[2 * 4];
+  {
+    const int drow = wave * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int ri = 0; ri < 4; ++ri) {
+      w3r[ri] = __bfloat162float(L.pol.w3[0][drow + ri]);
+      w3r[4 + ri] = __bfloat162float(L.pol.w3[1][drow + ri]);
+    }
+  }
 
   // prologue "phase A" for t=0: stats + normalized obs from the init state
   {
@@ -247,23 +308,9 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     mfma_strip_tanh<KPAD, S1, S1, S2>(&L.pol.w1[0][0], &L.xb[0][0],
                                       L.pol.b1, &L.h1[0][0], wave, lane);
     __syncthreads();
-    // ---- phase C: h2 = tanh(W2 h1 + b2) (MFMA, K=64) ------------------
-    mfma_strip_tanh<HID, S2, S2, S2>(&L.pol.w2[0][0], &L.h1[0][0],
-                                     L.pol.b2, &L.h2[0][0], wave, lane);
-    __syncthreads();
-    // ---- phase D1: logits partials (thread (env, quarter)) ------------
-    {
-      float p0 = 0.f, p1 = 0.f;
-#pragma unroll
-      for (int hh = 0; hh < HID / 4; ++hh) {
-        const int h = dd * (HID / 4) + hh;
-        const float hv = __bfloat162float(L.h2[env][h]);
-        p0 += __bfloat162float(L.pol.w3[0][h]) * hv;
-        p1 += __bfloat162float(L.pol.w3[1][h]) * hv;
-      }
-      L.lpart[dd][env][0] = p0;
-      L.lpart[dd][env][1] = p1;
-    }
+    // ---- phase C: h2 stays in registers; logits fused (MFMA, K=64) ----
+    mfma_strip_logits<HID, S2, S2>(&L.pol.w2[0][0], &L.h1[0][0],
+                                   L.pol.b2, w3r, L.lpart, wave, lane);
     __syncthreads();
     // ---- phase D2 (+A of t+1): action, env step, stats, next xb -------
     {

@@ -236,7 +236,9 @@ def rollout_reference(theta, sigma, seed, iteration, horizon, members,
         for _ in range(horizon):
             x = ((s - obs_mu) * rstd).clamp(-5, 5)
             h1 = bf(torch.tanh(bf(x) @ w1.T + b1))
-            h2 = bf(torch.tanh(h1 @ w2.T + b2))
+            # layer-2 activations stay fp32 in the kernel (fused-logits
+            # epilogue, no bf16 store)
+            h2 = torch.tanh(h1 @ w2.T + b2)
             logits = h2 @ w3.T + b3
             asign = torch.where(logits[:, 1] > logits[:, 0], 1.0, -1.0)
             drive = torch.tanh(s @ env_A.T)
