@@ -64,6 +64,74 @@ class Inventory:
         self._jobs = {}
         self._seq = itertools.count()
 
+    # -- dynamic (streaming imap) jobs: total size unknown until the
+    # feeder exhausts the iterator; results are freed as consumed -------
+    def add_dynamic(self, ordered):
+        seq = next(self._seq)
+        with self._lock:
+            self._jobs[seq] = {
+                "dyn": True,
+                "ordered": ordered,
+                "n": None,           # set when the iterator is exhausted
+                "results": {},       # ordered mode: index -> value
+                "queue": [],         # unordered mode: FIFO of values
+                "arrived": set(),    # dedup for resubmitted chunks
+                "yielded": 0,
+                "error": None,
+            }
+        return seq
+
+    def finish_dynamic(self, seq, n):
+        with self._cond:
+            job = self._jobs.get(seq)
+            if job is not None:
+                job["n"] = n
+            self._cond.notify_all()
+
+    def fail_dynamic(self, seq, exc):
+        with self._cond:
+            job = self._jobs.get(seq)
+            if job is not None and job["error"] is None:
+                job["error"] = _ExcInfo(exc)
+            self._cond.notify_all()
+
+    def dynamic_backlog(self, seq, base):
+        """Items submitted but not yet yielded (feeder back-pressure)."""
+        with self._lock:
+            job = self._jobs.get(seq)
+            if job is None:
+                return 0
+            return base - job["yielded"]
+
+    def iget_dynamic(self, seq, ordered):
+        index = 0
+        while True:
+            with self._cond:
+                job = self._jobs.get(seq)
+                if job is None:
+                    return
+                while True:
+                    if ordered and index in job["results"]:
+                        value = job["results"].pop(index)
+                        break
+                    if not ordered and job["queue"]:
+                        value = job["queue"].pop(0)
+                        break
+                    if job["error"] is not None:
+                        del self._jobs[seq]
+                        raise job["error"].rebuild()
+                    if job["n"] is not None and job["yielded"] >= job["n"]:
+                        del self._jobs[seq]
+                        return
+                    self._cond.wait()
+                job["yielded"] += 1
+            if isinstance(value, _ExcInfo):
+                with self._lock:
+                    self._jobs.pop(seq, None)
+                raise value.rebuild()
+            yield value
+            index += 1
+
     def add(self, n):
         seq = next(self._seq)
         with self._lock:
@@ -83,6 +151,27 @@ class Inventory:
             job = self._jobs.get(seq)
             if job is None:
                 return
+            if job.get("dyn"):
+                for offset, value in enumerate(values):
+                    index = base + offset
+                    if index in job["arrived"]:
+                        continue
+                    job["arrived"].add(index)
+                    if job["ordered"]:
+                        job["results"][index] = value
+                    else:
+                        job["queue"].append(value)
+                if failure is not None:
+                    index = base + failure[0]  # failure index is chunk-local
+                    exc_info = failure[1]
+                    if index not in job["arrived"]:
+                        job["arrived"].add(index)
+                        if job["ordered"]:
+                            job["results"][index] = exc_info
+                        else:
+                            job["queue"].append(exc_info)
+                self._cond.notify_all()
+                return
             for offset, value in enumerate(values):
                 index = base + offset
                 if job["arrived"][index]:
@@ -92,7 +181,8 @@ class Inventory:
                 job["remaining"] -= 1
                 job["unordered"].append((index, value))
             if failure is not None:
-                index, exc_info = failure
+                index = base + failure[0]  # failure index is chunk-local
+                exc_info = failure[1]
                 if not job["arrived"][index]:
                     job["arrived"][index] = True
                     job["remaining"] -= 1
@@ -625,12 +715,49 @@ class ZPool:
         )
 
     def imap(self, func, iterable, chunksize=1):
-        result = self._submit(func, iterable, chunksize, False)
-        return self._inventory.iget_ordered(result._seq)
+        return self._imap_lazy(func, iterable, chunksize, ordered=True)
 
     def imap_unordered(self, func, iterable, chunksize=1):
-        result = self._submit(func, iterable, chunksize, False)
-        return self._inventory.iget_unordered(result._seq)
+        return self._imap_lazy(func, iterable, chunksize, ordered=False)
+
+    def _imap_lazy(self, func, iterable, chunksize, ordered):
+        """Streaming imap (stdlib mp.Pool fidelity): a feeder thread pulls
+        from the iterator with back-pressure instead of materializing it;
+        results are freed as the consumer yields them."""
+        self._check_running()
+        self._lazy_start_workers(func)
+        iterator = iter(iterable)
+        seq = self._inventory.add_dynamic(ordered)
+        func_blob = serialization.dumps_closure(func)
+        window = max(256, 4 * chunksize * self._processes)
+
+        def feeder():
+            base = 0
+            try:
+                while True:
+                    if self._state != "run":
+                        self._inventory.fail_dynamic(
+                            seq, RuntimeError("pool closed during imap")
+                        )
+                        return
+                    if self._inventory.dynamic_backlog(seq, base) > window:
+                        time.sleep(0.002)
+                        continue
+                    chunk = list(itertools.islice(iterator, chunksize))
+                    if not chunk:
+                        self._inventory.finish_dynamic(seq, base)
+                        return
+                    self._taskq.put(
+                        (seq, base, func_blob, chunk, False, None)
+                    )
+                    base += len(chunk)
+            except Exception as exc:  # noqa: BLE001 (iterator may raise)
+                self._inventory.fail_dynamic(seq, exc)
+
+        threading.Thread(
+            target=feeder, name="fam-imap-feeder", daemon=True
+        ).start()
+        return self._inventory.iget_dynamic(seq, ordered)
 
     # -- lifecycle ---------------------------------------------------------
     def close(self):

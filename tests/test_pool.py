@@ -193,3 +193,88 @@ class TestPoolInitializer:
         finally:
             p.terminate()
             p.join()
+
+
+def _slow_square(x):
+    time.sleep(0.001)
+    return x * x
+
+
+class TestLazyImap:
+    def test_imap_streams_from_generator(self):
+        """imap must not materialize the iterable (stdlib fidelity):
+        feed a generator that tracks its own progress."""
+        pool = ZPool(processes=2)
+        pulled = []
+
+        def gen():
+            for i in range(5000):
+                pulled.append(i)
+                yield i
+
+        try:
+            it = pool.imap(_square, gen(), chunksize=4)
+            first = [next(it) for _ in range(8)]
+            assert first == [x * x for x in range(8)]
+            # back-pressure: the feeder must NOT have drained all 5000
+            assert len(pulled) < 5000
+            rest = list(it)
+            assert len(rest) == 5000 - 8
+            assert rest[-1] == 4999 * 4999
+            assert len(pulled) == 5000
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_imap_unordered_streams(self):
+        pool = ZPool(processes=4)
+        try:
+            got = sorted(pool.imap_unordered(_slow_square, iter(range(200)),
+                                             chunksize=2))
+            assert got == sorted(x * x for x in range(200))
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_imap_propagates_iterator_error(self):
+        pool = ZPool(processes=2)
+
+        def bad_gen():
+            yield 1
+            raise RuntimeError("iterator exploded")
+
+        try:
+            it = pool.imap(_square, bad_gen())
+            with pytest.raises(RuntimeError):
+                list(it)
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_imap_task_error_at_position(self):
+        pool = ZPool(processes=2)
+
+        def maybe_boom(x):
+            if x == 7:
+                raise ValueError("x7")
+            return x
+
+        try:
+            it = pool.imap(maybe_boom, range(20), chunksize=3)
+            got = [next(it) for _ in range(7)]
+            assert got == list(range(7))
+            with pytest.raises(ValueError):
+                next(it)
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_resilient_imap_chaos(self):
+        pool = ResilientZPool(processes=4)
+        try:
+            got = sorted(pool.imap_unordered(_random_error, iter(range(150)),
+                                             chunksize=2))
+            assert got == sorted(x * 2 for x in range(150))
+        finally:
+            pool.terminate()
+            pool.join()
