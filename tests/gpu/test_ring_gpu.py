@@ -17,10 +17,10 @@ requires_gpu = pytest.mark.skipif(
 @requires_gpu
 class TestRcclWorldOne:
     def _ctx(self):
-        from fiber_amd.ring import RingContext
+        from fiber_amd.ring import RingContext, _free_tcp_port
 
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29611")
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(_free_tcp_port())
         os.environ.setdefault("RANK", "0")
         os.environ.setdefault("WORLD_SIZE", "1")
         return RingContext(0, 1, backend="nccl",
