@@ -36,10 +36,16 @@ __device__ __host__ inline fam_uint4 philox4x32_r(uint32_t k0, uint32_t k1,
                                                   uint32_t c2, uint32_t c3) {
 #pragma unroll
   for (int r = 0; r < ROUNDS; ++r) {
-    uint32_t hi0 = fam_mulhi(FAM_PHILOX_M0, c0);
-    uint32_t lo0 = FAM_PHILOX_M0 * c0;
-    uint32_t hi1 = fam_mulhi(FAM_PHILOX_M1, c2);
-    uint32_t lo1 = FAM_PHILOX_M1 * c2;
+    // Single 64-bit multiplies: the GPU backend lowers these to one
+    // v_mad_u64_u32 each (hi+lo together) instead of a v_mul_hi_u32 +
+    // v_mul_lo_u32 pair — the mulhi chains dominate the philox-bound
+    // kernels (obsgen PMC: 102% VALUBusy).
+    uint64_t p0 = (uint64_t)FAM_PHILOX_M0 * c0;
+    uint64_t p1 = (uint64_t)FAM_PHILOX_M1 * c2;
+    uint32_t hi0 = (uint32_t)(p0 >> 32);
+    uint32_t lo0 = (uint32_t)p0;
+    uint32_t hi1 = (uint32_t)(p1 >> 32);
+    uint32_t lo1 = (uint32_t)p1;
     uint32_t n0 = hi1 ^ c1 ^ k0;
     uint32_t n1 = lo1;
     uint32_t n2 = hi0 ^ c3 ^ k1;
