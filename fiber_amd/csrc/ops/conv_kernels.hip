@@ -148,15 +148,19 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
   float s[SDIM];
 #pragma unroll
   for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
-  // each position needs CIN=4 values (one philox draw); two adjacent
-  // positions per thread iteration -> one 8 B store
-  for (int p = threadIdx.x * 2; p < IMG * IMG; p += blockDim.x * 2) {
+  // each position needs CIN=4 values (one philox draw); FOUR adjacent
+  // positions per thread iteration -> 4 independent philox chains in
+  // flight (the 10-round mad_u64 chain is serial, so ILP across
+  // positions is what fills the pipe) and one 16 B store.
+  // IMG*IMG = 7056 is a multiple of 4, so the strided 4-position loop
+  // covers every position exactly once (thread t owns p ≡ 4t mod 4096).
+  for (int p = threadIdx.x * 4; p + 3 < IMG * IMG; p += blockDim.x * 4) {
     union {
-      unsigned char b[8];
-      unsigned long long u;
+      unsigned char b[16];
+      uint32_t w[4];
     } pk;
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
+    for (int q = 0; q < 4; ++q) {
       float z[4];
       fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)(p + q), FAM_TAG_OBS,
                    t, z);
@@ -165,7 +169,8 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
       for (int c = 0; c < CIN; ++c)
         pk.b[q * 4 + c] = __hip_fp8_e4m3(0.52f * z[c] + s[c] * g).__x;
     }
-    *reinterpret_cast<unsigned long long*>(&out[p * CIN]) = pk.u;
+    *reinterpret_cast<fam_uint4*>(&out[p * CIN]) =
+        *reinterpret_cast<fam_uint4*>(pk.w);
   }
 }
 
