@@ -25,7 +25,7 @@ def leak_check():
     # for SIGTERM delivery on just-killed workers).
     import time
 
-    deadline = time.monotonic() + 3.0
+    deadline = time.monotonic() + 10.0
     leftover = fiber_amd.active_children()
     while leftover and time.monotonic() < deadline:
         time.sleep(0.05)
