@@ -89,3 +89,20 @@ class TestConvPipeline:
         f2 = eng.rollout(5).clone()
         torch.cuda.synchronize()
         assert torch.equal(f1, f2)
+
+
+@requires_gpu
+class TestGraphEquivalence:
+    def test_graph_replay_matches_eager(self):
+        """The hipGraph-replayed rollout must produce bitwise-identical
+        fitness to eager kernel launches."""
+        device = torch.device("cuda")
+        cfg = ConvESConfig(pop_per_gpu=16, horizon=6)
+        eager = ConvESEngine(cfg, ctx=None, device=device)
+        eager.use_graph = False
+        f_eager = eager.rollout(2).clone()
+        graphed = ConvESEngine(cfg, ctx=None, device=device)
+        f_graph = graphed.rollout(2).clone()
+        torch.cuda.synchronize()
+        assert graphed._graph is not None, "graph capture did not engage"
+        assert torch.equal(f_eager, f_graph)
