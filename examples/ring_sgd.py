@@ -18,7 +18,23 @@ import argparse
 import functools
 
 
-def train(rank, size, backend=None, steps=20):
+def make_convnet():
+    """The reference example's MNIST ConvNet (uber/fiber
+    examples/ring.py:89-105: conv 20@5x5 -> conv 50@5x5 -> fc500 -> 10)."""
+    import torch
+
+    return torch.nn.Sequential(
+        torch.nn.Conv2d(1, 20, 5), torch.nn.MaxPool2d(2),
+        torch.nn.ReLU(),
+        torch.nn.Conv2d(20, 50, 5), torch.nn.MaxPool2d(2),
+        torch.nn.ReLU(),
+        torch.nn.Flatten(),
+        torch.nn.Linear(800, 500), torch.nn.ReLU(),
+        torch.nn.Linear(500, 10),
+    )
+
+
+def train(rank, size, backend=None, steps=20, batch=64):
     import torch
 
     from fiber_amd.ring import RingContext
@@ -28,16 +44,14 @@ def train(rank, size, backend=None, steps=20):
     device = ctx.device
 
     torch.manual_seed(1234)  # identical init on every rank
-    model = torch.nn.Sequential(
-        torch.nn.Linear(64, 128), torch.nn.Tanh(),
-        torch.nn.Linear(128, 10),
-    ).to(device)
+    model = make_convnet().to(device)
     opt = torch.optim.SGD(model.parameters(), lr=0.05)
 
     gen = torch.Generator().manual_seed(rank)  # rank-local shard
     for step in range(steps):
-        x = torch.randn(256, 64, generator=gen).to(device)
-        y = (x.sum(dim=1) > 0).long().to(device)
+        # synthetic MNIST-shaped data with a learnable rule
+        x = torch.randn(batch, 1, 28, 28, generator=gen).to(device)
+        y = (x.mean(dim=(1, 2, 3)) > 0).long().to(device)
         loss = torch.nn.functional.cross_entropy(model(x), y)
         opt.zero_grad()
         loss.backward()
