@@ -278,3 +278,28 @@ class TestLazyImap:
         finally:
             pool.terminate()
             pool.join()
+
+
+def _inner_times10(x):
+    return x * 10
+
+
+def _nested_outer(n):
+    pool = ZPool(processes=2)
+    try:
+        return sum(pool.map(_inner_times10, range(n)))
+    finally:
+        pool.terminate()
+        pool.join()
+
+
+class TestNestedPools:
+    def test_worker_creates_its_own_pool(self):
+        """Nested fiber processes (reference-supported pattern): a pool
+        worker spins up its own pool."""
+        pool = ZPool(processes=2)
+        try:
+            assert pool.map(_nested_outer, [3, 4]) == [30, 60]
+        finally:
+            pool.terminate()
+            pool.join()
