@@ -350,19 +350,17 @@ conv_fc(const __hip_bfloat16* __restrict__ wpert,
   const unsigned char* ap = &w3[(size_t)arow * NFLAT + kgrp * 8];
   const unsigned char* bp = &in[(size_t)env * NFLAT + kgrp * 8];
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int kk = 0; kk < 80; kk += 4) {
-    const fp8x8 a0 = *reinterpret_cast<const fp8x8*>(ap + (kk + 0) * 32);
-    const fp8x8 b0 = *reinterpret_cast<const fp8x8*>(bp + (kk + 0) * 32);
-    const fp8x8 a1 = *reinterpret_cast<const fp8x8*>(ap + (kk + 1) * 32);
-    const fp8x8 b1 = *reinterpret_cast<const fp8x8*>(bp + (kk + 1) * 32);
-    const fp8x8 a2 = *reinterpret_cast<const fp8x8*>(ap + (kk + 2) * 32);
-    const fp8x8 b2 = *reinterpret_cast<const fp8x8*>(bp + (kk + 2) * 32);
-    const fp8x8 a3 = *reinterpret_cast<const fp8x8*>(ap + (kk + 3) * 32);
-    const fp8x8 b3 = *reinterpret_cast<const fp8x8*>(bp + (kk + 3) * 32);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0, 0);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc, 0, 0, 0);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b2, acc, 0, 0, 0);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b3, acc, 0, 0, 0);
+  for (int kk = 0; kk < 80; kk += 8) {
+    fp8x8 a[8], b[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      a[u] = *reinterpret_cast<const fp8x8*>(ap + (kk + u) * 32);
+      b[u] = *reinterpret_cast<const fp8x8*>(bp + (kk + u) * 32);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a[u], b[u], acc, 0,
+                                                       0, 0);
   }
   {  // tail (81st K-tile)
     const fp8x8 a = *reinterpret_cast<const fp8x8*>(ap + 80 * 32);
