@@ -195,6 +195,10 @@ class Inventory:
         """Abort every outstanding job (pool terminated)."""
         with self._cond:
             for job in self._jobs.values():
+                if job.get("dyn"):
+                    if job["error"] is None:
+                        job["error"] = _ExcInfo(exc)
+                    continue
                 if job["remaining"] > 0:
                     job["error"] = _ExcInfo(exc)
                     job["remaining"] = 0
@@ -204,6 +208,16 @@ class Inventory:
         with self._lock:
             job = self._jobs.get(seq)
             return job is None or job["remaining"] == 0
+
+    def peek(self, seq):
+        """(done, error, results) without consuming (callback watcher)."""
+        with self._lock:
+            job = self._jobs.get(seq)
+            if job is None:
+                return True, None, None
+            done = job["remaining"] == 0 or job["error"] is not None
+            results = list(job["results"]) if done else None
+            return done, job["error"], results
 
     def get(self, seq, timeout=None):
         deadline = None if timeout is None else time.monotonic() + timeout
@@ -271,18 +285,34 @@ class AsyncResult:
         self._error_callback = error_callback
 
     def get(self, timeout=None):
+        # callbacks fire from the pool's result thread on completion
+        # (mp.Pool semantics), not here
         results = self._pool._inventory.get(self._seq, timeout)
         first_exc = next(
             (r for r in results if isinstance(r, _ExcInfo)), None
         )
         if first_exc is not None:
-            if self._error_callback:
-                self._error_callback(first_exc.rebuild())
             raise first_exc.rebuild()
-        out = results[0] if self._single else results
-        if self._callback:
-            self._callback(out)
-        return out
+        return results[0] if self._single else results
+
+    def _fire_callbacks(self, error, results):
+        """Called once by the pool's result thread when the job is done."""
+        try:
+            first_exc = error or next(
+                (r for r in results if isinstance(r, _ExcInfo)), None
+            )
+            if first_exc is not None:
+                if self._error_callback:
+                    self._error_callback(first_exc.rebuild())
+            elif self._callback:
+                self._callback(results[0] if self._single else results)
+        except Exception:  # noqa: BLE001 - callback errors must not kill
+            import traceback
+
+            traceback.print_exc()
+        finally:
+            self._callback = None
+            self._error_callback = None
 
     def wait(self, timeout=None):
         try:
@@ -500,6 +530,7 @@ class ZPool:
         self._inventory = Inventory()
         self._taskq = _stdlib_queue.Queue()
         self._workers = {}  # ident -> Process
+        self._callback_watch = {}  # seq -> AsyncResult with callbacks
         self._all_idents = set()
         self._worker_lock = threading.Lock()
         self._state = "run"  # run -> closing -> terminated
@@ -635,6 +666,12 @@ class ZPool:
             self._recv += 1
             self._ack(ident, seq, base)
             self._inventory.put(seq, base, values, failure)
+            watcher = self._callback_watch.get(seq)
+            if watcher is not None:
+                done, error, results = self._inventory.peek(seq)
+                if done:
+                    self._callback_watch.pop(seq, None)
+                    watcher._fire_callbacks(error, results)
 
     def _ack(self, ident, seq, base):
         pass  # resilient subclass clears pending table
@@ -657,7 +694,11 @@ class ZPool:
         )
         if n == 0:
             self._inventory.put(seq, 0, [], None)
+            if callback is not None:
+                result._fire_callbacks(None, [])
             return result
+        if callback is not None or error_callback is not None:
+            self._callback_watch[seq] = result
         func_blob = serialization.dumps_closure(func)
         if chunksize is None:
             chunksize = max(1, min(DEFAULT_CHUNKSIZE, n // 4 or 1))

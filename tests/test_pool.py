@@ -303,3 +303,53 @@ class TestNestedPools:
         finally:
             pool.terminate()
             pool.join()
+
+
+class TestAsyncCallbacks:
+    def test_callback_fires_without_get(self):
+        """mp.Pool semantics: callbacks fire from the pool when the job
+        completes, independent of .get()."""
+        pool = ZPool(processes=2)
+        seen = []
+        try:
+            pool.map_async(_square, range(10), callback=seen.append)
+            deadline = time.monotonic() + 30
+            while not seen and time.monotonic() < deadline:
+                time.sleep(0.01)
+            assert seen == [[x * x for x in range(10)]]
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_error_callback_fires(self):
+        pool = ZPool(processes=2)
+        errors = []
+
+        def boom(x):
+            raise ValueError("cb-%d" % x)
+
+        try:
+            r = pool.map_async(boom, range(4), chunksize=1,
+                               error_callback=errors.append)
+            deadline = time.monotonic() + 30
+            while not errors and time.monotonic() < deadline:
+                time.sleep(0.01)
+            assert errors and isinstance(errors[0], ValueError)
+            with pytest.raises(ValueError):
+                r.get(10)
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_apply_async_callback(self):
+        pool = ZPool(processes=2)
+        seen = []
+        try:
+            pool.apply_async(_add, (2, 3), callback=seen.append)
+            deadline = time.monotonic() + 30
+            while not seen and time.monotonic() < deadline:
+                time.sleep(0.01)
+            assert seen == [5]
+        finally:
+            pool.terminate()
+            pool.join()
