@@ -102,6 +102,30 @@ inline void make_deadline(double timeout_s, struct timespec* ts) {
   }
 }
 
+// cond waits re-acquire the mutex internally and can therefore return
+// EOWNERDEAD too (previous owner died while holding).  Ignoring that and
+// re-waiting on an inconsistent mutex is UB (observed as a permanent
+// wedge under worker-kill chaos) — every wait must run
+// pthread_mutex_consistent before continuing.
+inline int cond_wait_robust(pthread_cond_t* c, pthread_mutex_t* m) {
+  int rc = pthread_cond_wait(c, m);
+  if (rc == EOWNERDEAD) {
+    pthread_mutex_consistent(m);
+    rc = 0;
+  }
+  return rc;
+}
+
+inline int cond_timedwait_robust(pthread_cond_t* c, pthread_mutex_t* m,
+                                 const struct timespec* ts) {
+  int rc = pthread_cond_timedwait(c, m, ts);
+  if (rc == EOWNERDEAD) {
+    pthread_mutex_consistent(m);
+    rc = 0;
+  }
+  return rc;
+}
+
 class RobustLock {
  public:
   explicit RobustLock(pthread_mutex_t* mu) : mu_(mu) {
@@ -245,10 +269,10 @@ class ShmRing {
         if (timeout == 0) return false;
         int rc;
         if (timeout < 0) {
-          rc = pthread_cond_wait(&hdr_->not_full, &hdr_->mu);
+          rc = cond_wait_robust(&hdr_->not_full, &hdr_->mu);
         } else {
-          rc = pthread_cond_timedwait(&hdr_->not_full, &hdr_->mu,
-                                      &deadline);
+          rc = cond_timedwait_robust(&hdr_->not_full, &hdr_->mu,
+                                     &deadline);
           if (rc == ETIMEDOUT) return false;
         }
         (void)rc;
@@ -342,7 +366,7 @@ class ShmRing {
         }
         // short poll wait for the commit
         make_deadline(0.1, &tick);
-        int rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        int rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
         if (rc == ETIMEDOUT) {
           ++stuck_polls;
           if (timeout > 0) {
@@ -361,9 +385,9 @@ class ShmRing {
       if (timeout == 0) return false;
       int rc;
       if (timeout < 0) {
-        rc = pthread_cond_wait(&hdr_->not_empty, &hdr_->mu);
+        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
       } else {
-        rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &deadline);
+        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
         if (rc == ETIMEDOUT) return false;
       }
       (void)rc;
@@ -393,7 +417,7 @@ class ShmRing {
                 ->load(std::memory_order_acquire);
         if (state == kStCommitted) return (int64_t)len32;
         make_deadline(0.1, &tick);
-        pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
         if (timeout == 0) return -1;
         continue;
       }
@@ -401,9 +425,9 @@ class ShmRing {
       if (timeout == 0) return -1;
       int rc;
       if (timeout < 0) {
-        rc = pthread_cond_wait(&hdr_->not_empty, &hdr_->mu);
+        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
       } else {
-        rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &deadline);
+        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
         if (rc == ETIMEDOUT) return -1;
       }
       (void)rc;
@@ -459,7 +483,7 @@ class ShmRing {
           stuck_polls = 0;
         }
         make_deadline(0.1, &tick);
-        int rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &tick);
+        int rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
         if (rc == ETIMEDOUT) {
           ++stuck_polls;
           if (timeout == 0) return -1;
@@ -478,9 +502,9 @@ class ShmRing {
       if (timeout == 0) return -1;
       int rc;
       if (timeout < 0) {
-        rc = pthread_cond_wait(&hdr_->not_empty, &hdr_->mu);
+        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
       } else {
-        rc = pthread_cond_timedwait(&hdr_->not_empty, &hdr_->mu, &deadline);
+        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
         if (rc == ETIMEDOUT) return -1;
       }
       (void)rc;
