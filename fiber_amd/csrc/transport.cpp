@@ -513,10 +513,22 @@ class ShmRing {
 
   void close_ring() {
     if (!hdr_) return;
-    RobustLock lock(&hdr_->mu);
+    // Time-boxed: teardown must never wedge even if the mutex is stuck
+    // (e.g. an undiagnosed robust-mutex corner under SIGKILL chaos).
+    // The closed flag is a u32 checked inside every wait loop, so a
+    // lock-free store + broadcast still wakes and drains all waiters.
+    struct timespec deadline;
+    clock_gettime(CLOCK_REALTIME, &deadline);
+    deadline.tv_sec += 5;
+    int rc = pthread_mutex_timedlock(&hdr_->mu, &deadline);
+    if (rc == EOWNERDEAD) {
+      pthread_mutex_consistent(&hdr_->mu);
+      rc = 0;
+    }
     hdr_->closed = 1;
     pthread_cond_broadcast(&hdr_->not_empty);
     pthread_cond_broadcast(&hdr_->not_full);
+    if (rc == 0) pthread_mutex_unlock(&hdr_->mu);
   }
 
   void unlink_ring() { shm_unlink(name_.c_str()); }
