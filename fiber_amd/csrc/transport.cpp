@@ -2,21 +2,35 @@
 //
 // Role: replaces the reference's nanomsg TCP data plane (uber/fiber
 // fiber/socket.py:297-425 NanomsgContext/NanomsgDevice) with an MI355X-node
-// native engine: variable-size message rings in POSIX shared memory guarded
-// by process-shared ROBUST pthread mutexes + condvars.  A ring is MPMC:
-// any number of producers and consumers in any process on the node.  A
-// consumer that dies while holding the lock does not deadlock the ring
-// (EOWNERDEAD -> pthread_mutex_consistent), which is what makes the
-// resilient pool's worker-kill recovery safe at the transport level.
+// native engine: variable-size message rings in POSIX shared memory.
+// A ring is MPMC: any number of producers and consumers in any process on
+// the node.
 //
-// Message payloads are opaque bytes (pickled host metadata; device tensors
-// ride as ~100-byte HIP IPC handles produced by fiber_amd.serialization,
-// so the ring never carries tensor data).
+// CRASH SAFETY (the whole point of the design):
+//   * state mutations are guarded by a process-shared ROBUST pthread
+//     mutex — a process dying inside the critical section hands the next
+//     locker EOWNERDEAD + a consistent-enough state (head/tail/used are
+//     mutated only after payload/header writes);
+//   * blocking uses RAW FUTEX sequence words, NOT pthread condvars:
+//     glibc condvars contain an internal NON-robust lock, so a process
+//     SIGKILLed inside cond_wait's bookkeeping permanently wedges every
+//     later broadcast/wait (observed live under worker-kill chaos).
+//     Futex wait/wake are single syscalls with no shared-state locking —
+//     kill-safe at every instruction.  All waits are additionally chunked
+//     at 1 s so even a missed wake (producer dying between its state
+//     update and its FUTEX_WAKE) only costs a bounded stall;
+//   * a payload is copied OUTSIDE the lock (reserve -> copy -> commit).
+//     A reserved record whose writer died (incl. zombies) is reclaimed
+//     by the next reader.
+//
+// Message payloads are opaque bytes (pickled host metadata; device
+// tensors ride as ~100-byte HIP IPC handles produced by
+// fiber_amd.serialization, so the ring never carries tensor data).
 //
 // Layout of a segment (/dev/shm/<name>):
 //   [Header | data area of `capacity` bytes]
-// Records in the data area: [u32 len][payload][pad to 8B].  A WRAP marker
-// (len == 0xFFFFFFFF) means "skip to offset 0".
+// Records: [u32 len][u32 state][u32 writer_pid][u32 pad][payload..] padded
+// to 8 B.  A WRAP marker (len == 0xFFFFFFFF) means "skip to offset 0".
 
 #include <pybind11/pybind11.h>
 
@@ -29,10 +43,12 @@
 #include <string>
 
 #include <fcntl.h>
+#include <linux/futex.h>
 #include <pthread.h>
+#include <signal.h>
 #include <sys/mman.h>
 #include <sys/stat.h>
-#include <signal.h>
+#include <sys/syscall.h>
 #include <time.h>
 #include <unistd.h>
 
@@ -40,14 +56,10 @@ namespace py = pybind11;
 
 namespace {
 
-constexpr uint32_t kMagic = 0xFA3B71A6u;
+constexpr uint32_t kMagic = 0xFA3B71A7u;  // bumped: futex-word header
 constexpr uint32_t kWrapMarker = 0xFFFFFFFFu;
 constexpr size_t kAlign = 8;
 
-// Record layout: [u32 len][u32 state][u32 writer_pid][u32 _pad][payload..]
-// state: 0 = reserved (payload being copied outside the lock),
-//        1 = committed.  A reserved record whose writer died is reclaimed
-// by the next reader (crash safety for mid-copy kills).
 constexpr uint32_t kStReserved = 0u;
 constexpr uint32_t kStCommitted = 1u;
 constexpr size_t kRecHdr = 16;
@@ -56,8 +68,10 @@ struct Header {
   uint32_t magic;
   uint32_t ready;  // set to 1 once the creator finished initialization
   pthread_mutex_t mu;
-  pthread_cond_t not_empty;
-  pthread_cond_t not_full;
+  // futex sequence words: bumped (under the mutex) whenever the
+  // corresponding condition may have become true, woken with FUTEX_WAKE
+  uint32_t fut_not_empty;
+  uint32_t fut_not_full;
   uint64_t capacity;   // bytes in data area
   uint64_t head;       // consumer offset
   uint64_t tail;       // producer offset
@@ -90,40 +104,36 @@ inline bool process_alive(uint32_t pid) {
   return true;
 }
 
-inline void make_deadline(double timeout_s, struct timespec* ts) {
-  clock_gettime(CLOCK_MONOTONIC, ts);
-  time_t sec = static_cast<time_t>(timeout_s);
-  long nsec = static_cast<long>((timeout_s - (double)sec) * 1e9);
-  ts->tv_sec += sec;
-  ts->tv_nsec += nsec;
-  if (ts->tv_nsec >= 1000000000L) {
-    ts->tv_sec += 1;
-    ts->tv_nsec -= 1000000000L;
-  }
+inline double monotonic_now() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return ts.tv_sec + ts.tv_nsec * 1e-9;
 }
 
-// cond waits re-acquire the mutex internally and can therefore return
-// EOWNERDEAD too (previous owner died while holding).  Ignoring that and
-// re-waiting on an inconsistent mutex is UB (observed as a permanent
-// wedge under worker-kill chaos) — every wait must run
-// pthread_mutex_consistent before continuing.
-inline int cond_wait_robust(pthread_cond_t* c, pthread_mutex_t* m) {
-  int rc = pthread_cond_wait(c, m);
-  if (rc == EOWNERDEAD) {
-    pthread_mutex_consistent(m);
-    rc = 0;
-  }
-  return rc;
+inline std::atomic<uint32_t>* as_atomic(uint32_t* p) {
+  return reinterpret_cast<std::atomic<uint32_t>*>(p);
 }
 
-inline int cond_timedwait_robust(pthread_cond_t* c, pthread_mutex_t* m,
-                                 const struct timespec* ts) {
-  int rc = pthread_cond_timedwait(c, m, ts);
-  if (rc == EOWNERDEAD) {
-    pthread_mutex_consistent(m);
-    rc = 0;
-  }
-  return rc;
+// Shared (cross-process) futex wait: returns after a wake, timeout,
+// value mismatch or spurious wakeup — callers always re-check state.
+inline void futex_wait(uint32_t* word, uint32_t expected, double max_s) {
+  struct timespec rel;
+  if (max_s > 1.0) max_s = 1.0;  // bounded: missed-wake insurance
+  if (max_s <= 0) return;
+  rel.tv_sec = (time_t)max_s;
+  rel.tv_nsec = (long)((max_s - (double)rel.tv_sec) * 1e9);
+  syscall(SYS_futex, word, FUTEX_WAIT, expected, &rel, nullptr, 0);
+}
+
+inline void futex_wake_all(uint32_t* word) {
+  syscall(SYS_futex, word, FUTEX_WAKE, INT32_MAX, nullptr, nullptr, 0);
+}
+
+// Bump the sequence word (must hold the mutex so waiters' snapshot-then-
+// wait ordering is race-free) and wake everyone.
+inline void bump_and_wake(uint32_t* word) {
+  as_atomic(word)->fetch_add(1, std::memory_order_release);
+  futex_wake_all(word);
 }
 
 class RobustLock {
@@ -132,14 +142,20 @@ class RobustLock {
     int rc = pthread_mutex_lock(mu_);
     if (rc == EOWNERDEAD) {
       // Previous owner died mid-critical-section.  Ring mutations are
-      // ordered so that head/tail/used advance only after payload writes,
-      // so the state is consistent enough to continue.
+      // ordered so that head/tail/used advance only after payload
+      // writes, so the state is consistent enough to continue.
       pthread_mutex_consistent(mu_);
     } else if (rc != 0) {
       throw std::runtime_error("mutex lock failed: " + std::to_string(rc));
     }
   }
-  ~RobustLock() { pthread_mutex_unlock(mu_); }
+  ~RobustLock() {
+    if (mu_) pthread_mutex_unlock(mu_);
+  }
+  void unlock() {
+    pthread_mutex_unlock(mu_);
+    mu_ = nullptr;
+  }
 
  private:
   pthread_mutex_t* mu_;
@@ -155,7 +171,8 @@ class ShmRing {
     if (create) {
       shm_unlink(name.c_str());  // stale segment from a crashed run
       fd = shm_open(name.c_str(), O_CREAT | O_EXCL | O_RDWR, 0600);
-      if (fd < 0) throw std::runtime_error("shm_open create failed: " + name);
+      if (fd < 0)
+        throw std::runtime_error("shm_open create failed: " + name);
       if (ftruncate(fd, (off_t)total) != 0) {
         close(fd);
         shm_unlink(name.c_str());
@@ -163,16 +180,11 @@ class ShmRing {
       }
     } else {
       // The binder may not have created the segment yet; retry briefly.
-      struct timespec start;
-      clock_gettime(CLOCK_MONOTONIC, &start);
+      double start = monotonic_now();
       for (;;) {
         fd = shm_open(name.c_str(), O_RDWR, 0600);
         if (fd >= 0) break;
-        struct timespec now;
-        clock_gettime(CLOCK_MONOTONIC, &now);
-        double waited = (now.tv_sec - start.tv_sec) +
-                        (now.tv_nsec - start.tv_nsec) * 1e-9;
-        if (waited > open_timeout)
+        if (monotonic_now() - start > open_timeout)
           throw std::runtime_error("shm ring not found: " + name);
         usleep(2000);
       }
@@ -188,8 +200,7 @@ class ShmRing {
     void* mem =
         mmap(nullptr, total, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
     close(fd);
-    if (mem == MAP_FAILED)
-      throw std::runtime_error("mmap failed: " + name);
+    if (mem == MAP_FAILED) throw std::runtime_error("mmap failed: " + name);
     hdr_ = reinterpret_cast<Header*>(mem);
     data_ = reinterpret_cast<char*>(mem) + sizeof(Header);
     map_len_ = total;
@@ -205,27 +216,13 @@ class ShmRing {
       pthread_mutex_init(&hdr_->mu, &ma);
       pthread_mutexattr_destroy(&ma);
 
-      pthread_condattr_t ca;
-      pthread_condattr_init(&ca);
-      pthread_condattr_setpshared(&ca, PTHREAD_PROCESS_SHARED);
-      pthread_condattr_setclock(&ca, CLOCK_MONOTONIC);
-      pthread_cond_init(&hdr_->not_empty, &ca);
-      pthread_cond_init(&hdr_->not_full, &ca);
-      pthread_condattr_destroy(&ca);
-
       hdr_->magic = kMagic;
       std::atomic_thread_fence(std::memory_order_release);
       hdr_->ready = 1;
     } else {
-      // Wait until creator finished header init.
-      struct timespec start;
-      clock_gettime(CLOCK_MONOTONIC, &start);
+      double start = monotonic_now();
       while (hdr_->ready != 1) {
-        struct timespec now;
-        clock_gettime(CLOCK_MONOTONIC, &now);
-        double waited = (now.tv_sec - start.tv_sec) +
-                        (now.tv_nsec - start.tv_nsec) * 1e-9;
-        if (waited > open_timeout) {
+        if (monotonic_now() - start > open_timeout) {
           munmap(mem, total);
           throw std::runtime_error("shm ring never became ready: " + name);
         }
@@ -250,273 +247,194 @@ class ShmRing {
 
   // timeout < 0: block forever; timeout == 0: non-blocking.
   // Returns false on timeout; throws if the ring is closed.
-  // The payload memcpy happens OUTSIDE the lock (reserve -> copy ->
-  // commit), so concurrent producers/consumers overlap their copies.
   bool send(const char* buf, size_t len, double timeout) {
     size_t need = record_bytes(len);
     if (need + kRecHdr + kAlign >= hdr_->capacity)
       throw std::runtime_error("message larger than ring capacity");
-    struct timespec deadline;
-    if (timeout > 0) make_deadline(timeout, &deadline);
+    const double deadline =
+        timeout > 0 ? monotonic_now() + timeout : 0.0;
 
     uint64_t rec;
-    {
-      RobustLock lock(&hdr_->mu);
-      for (;;) {
+    for (;;) {
+      // snapshot BEFORE the predicate check: any state change after this
+      // bumps the word, so our futex_wait cannot sleep through it
+      uint32_t snap = as_atomic(&hdr_->fut_not_full)
+                          ->load(std::memory_order_acquire);
+      {
+        RobustLock lock(&hdr_->mu);
         if (hdr_->closed) throw std::runtime_error("ring closed");
         // Worst case we also need a wrap marker record.
-        if (hdr_->capacity - hdr_->used >= need + kRecHdr + kAlign) break;
-        if (timeout == 0) return false;
-        int rc;
-        if (timeout < 0) {
-          rc = cond_wait_robust(&hdr_->not_full, &hdr_->mu);
-        } else {
-          rc = cond_timedwait_robust(&hdr_->not_full, &hdr_->mu,
-                                     &deadline);
-          if (rc == ETIMEDOUT) return false;
+        if (hdr_->capacity - hdr_->used >= need + kRecHdr + kAlign) {
+          uint64_t cap = hdr_->capacity;
+          uint64_t tail = hdr_->tail;
+          if (tail + need > cap) {
+            uint32_t marker = kWrapMarker;
+            std::memcpy(data_ + tail, &marker, 4);
+            hdr_->used += cap - tail;
+            tail = 0;
+          }
+          rec = tail;
+          uint32_t len32 = (uint32_t)len;
+          uint32_t pid = (uint32_t)getpid();
+          std::memcpy(data_ + rec, &len32, 4);
+          as_atomic(reinterpret_cast<uint32_t*>(data_ + rec + 4))
+              ->store(kStReserved, std::memory_order_relaxed);
+          std::memcpy(data_ + rec + 8, &pid, 4);
+          hdr_->tail = (tail + need) % cap;
+          hdr_->used += need;
+          hdr_->msg_count += 1;
+          hdr_->total_in += 1;
+          break;
         }
-        (void)rc;
+        if (timeout == 0) return false;
       }
-
-      uint64_t cap = hdr_->capacity;
-      uint64_t tail = hdr_->tail;
-      if (tail + need > cap) {
-        // Not enough contiguous space: write wrap marker, jump to 0.
-        uint32_t marker = kWrapMarker;
-        std::memcpy(data_ + tail, &marker, 4);
-        hdr_->used += cap - tail;
-        tail = 0;
+      double remaining = 1.0;
+      if (timeout > 0) {
+        remaining = deadline - monotonic_now();
+        if (remaining <= 0) return false;
       }
-      rec = tail;
-      uint32_t len32 = (uint32_t)len;
-      uint32_t pid = (uint32_t)getpid();
-      std::memcpy(data_ + rec, &len32, 4);
-      reinterpret_cast<std::atomic<uint32_t>*>(data_ + rec + 4)
-          ->store(kStReserved, std::memory_order_relaxed);
-      std::memcpy(data_ + rec + 8, &pid, 4);
-      hdr_->tail = (tail + need) % cap;
-      hdr_->used += need;
-      hdr_->msg_count += 1;
-      hdr_->total_in += 1;
+      futex_wait(&hdr_->fut_not_full, snap, remaining);
     }
 
     if (len) std::memcpy(data_ + rec + kRecHdr, buf, len);
-    reinterpret_cast<std::atomic<uint32_t>*>(data_ + rec + 4)
+    as_atomic(reinterpret_cast<uint32_t*>(data_ + rec + 4))
         ->store(kStCommitted, std::memory_order_release);
-    {
-      // Lock-protected signal so a reader between predicate-check and
-      // cond_wait cannot miss the wakeup.
-      RobustLock lock(&hdr_->mu);
-      pthread_cond_signal(&hdr_->not_empty);
-    }
+    bump_and_wake(&hdr_->fut_not_empty);
     return true;
   }
 
-  // Returns (found, payload).  found=false on timeout.  Throws when the
-  // ring is closed AND drained.
+  // Consumer-side helper: with the lock held, resolve the head record.
+  // Returns: 0 = a committed record is ready (out params set);
+  //          1 = ring empty; 2 = head reserved (writer mid-copy).
+  int resolve_head(uint64_t* head_out, uint32_t* len_out) {
+    if (hdr_->msg_count == 0) return 1;
+    uint64_t cap = hdr_->capacity;
+    uint64_t head = hdr_->head;
+    uint32_t len32;
+    std::memcpy(&len32, data_ + head, 4);
+    if (len32 == kWrapMarker) {
+      hdr_->used -= cap - head;
+      hdr_->head = head = 0;
+      std::memcpy(&len32, data_ + head, 4);
+    }
+    uint32_t state =
+        as_atomic(reinterpret_cast<uint32_t*>(data_ + head + 4))
+            ->load(std::memory_order_acquire);
+    *head_out = head;
+    *len_out = len32;
+    return state == kStCommitted ? 0 : 2;
+  }
+
+  void consume_record(uint64_t head, uint32_t len32) {
+    size_t need = record_bytes(len32);
+    hdr_->head = (head + need) % hdr_->capacity;
+    hdr_->used -= need;
+    hdr_->msg_count -= 1;
+    hdr_->total_out += 1;
+  }
+
+  void reclaim_dead_record(uint64_t head, uint32_t len32) {
+    size_t need = record_bytes(len32);
+    hdr_->head = (head + need) % hdr_->capacity;
+    hdr_->used -= need;
+    hdr_->msg_count -= 1;
+  }
+
+  // Shared blocking structure for recv / recv_into / peek.
+  // op(head, len) -> true when it consumed / is satisfied.
+  template <typename Op>
+  bool recv_loop(double timeout, Op&& op) {
+    const double deadline =
+        timeout > 0 ? monotonic_now() + timeout : 0.0;
+    int reserved_streak = 0;
+    for (;;) {
+      uint32_t snap = as_atomic(&hdr_->fut_not_empty)
+                          ->load(std::memory_order_acquire);
+      {
+        RobustLock lock(&hdr_->mu);
+        uint64_t head;
+        uint32_t len32;
+        int st = resolve_head(&head, &len32);
+        if (st == 0) {
+          if (op(head, len32)) {
+            lock.unlock();
+            bump_and_wake(&hdr_->fut_not_full);
+            return true;
+          }
+          return false;  // op declined (recv_into: buffer too small)
+        }
+        if (st == 1 && hdr_->closed)
+          throw std::runtime_error("ring closed");
+        if (st == 2) {
+          // writer mid-copy; reclaim if it died (incl. zombie)
+          ++reserved_streak;
+          if (reserved_streak > 3) {
+            uint32_t pid;
+            std::memcpy(&pid, data_ + head + 8, 4);
+            if (!process_alive(pid)) {
+              reclaim_dead_record(head, len32);
+              lock.unlock();
+              bump_and_wake(&hdr_->fut_not_full);
+              reserved_streak = 0;
+              continue;
+            }
+            reserved_streak = 0;
+          }
+        } else {
+          reserved_streak = 0;
+        }
+        if (timeout == 0) return false;
+      }
+      double remaining = st2_wait_;
+      if (timeout > 0) {
+        remaining = deadline - monotonic_now();
+        if (remaining <= 0) return false;
+      }
+      futex_wait(&hdr_->fut_not_empty, snap, remaining);
+    }
+  }
+
   bool recv(std::string* out, double timeout) {
-    struct timespec deadline;
-    if (timeout > 0) make_deadline(timeout, &deadline);
-    // Bounded wait even for infinite timeouts so dead-writer reclaim
-    // gets a chance to run.
-    struct timespec tick;
-
-    RobustLock lock(&hdr_->mu);
-    int stuck_polls = 0;
-    for (;;) {
-      if (hdr_->msg_count > 0) {
-        // resolve head (following a wrap marker) and check commit state
-        uint64_t cap = hdr_->capacity;
-        uint64_t head = hdr_->head;
-        uint32_t len32;
-        std::memcpy(&len32, data_ + head, 4);
-        if (len32 == kWrapMarker) {
-          hdr_->used -= cap - head;
-          hdr_->head = head = 0;
-          std::memcpy(&len32, data_ + head, 4);
-        }
-        uint32_t state =
-            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
-                ->load(std::memory_order_acquire);
-        if (state == kStCommitted) {
-          out->assign(data_ + head + kRecHdr, len32);
-          size_t need = record_bytes(len32);
-          hdr_->head = (head + need) % cap;
-          hdr_->used -= need;
-          hdr_->msg_count -= 1;
-          hdr_->total_out += 1;
-          pthread_cond_signal(&hdr_->not_full);
-          return true;
-        }
-        // Reserved record: writer is copying.  If the writer died
-        // mid-copy, reclaim the record so the ring cannot wedge.
-        if (stuck_polls > 20) {  // ~2s of 100ms polls
-          uint32_t pid;
-          std::memcpy(&pid, data_ + head + 8, 4);
-          if (!process_alive(pid)) {
-            size_t need = record_bytes(len32);
-            hdr_->head = (head + need) % cap;
-            hdr_->used -= need;
-            hdr_->msg_count -= 1;
-            pthread_cond_signal(&hdr_->not_full);
-            stuck_polls = 0;
-            continue;
-          }
-          stuck_polls = 0;
-        }
-        // short poll wait for the commit
-        make_deadline(0.1, &tick);
-        int rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
-        if (rc == ETIMEDOUT) {
-          ++stuck_polls;
-          if (timeout > 0) {
-            struct timespec now;
-            clock_gettime(CLOCK_MONOTONIC, &now);
-            if (now.tv_sec > deadline.tv_sec ||
-                (now.tv_sec == deadline.tv_sec &&
-                 now.tv_nsec >= deadline.tv_nsec))
-              return false;
-          }
-          if (timeout == 0) return false;
-        }
-        continue;
-      }
-      if (hdr_->closed) throw std::runtime_error("ring closed");
-      if (timeout == 0) return false;
-      int rc;
-      if (timeout < 0) {
-        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
-      } else {
-        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
-        if (rc == ETIMEDOUT) return false;
-      }
-      (void)rc;
-    }
+    return recv_loop(timeout, [&](uint64_t head, uint32_t len32) {
+      out->assign(data_ + head + kRecHdr, len32);
+      consume_record(head, len32);
+      return true;
+    });
   }
 
-  // Size of the next committed message, or -1 on timeout.  Does not
-  // consume.  (Used with recv_into for single-copy receives.)
-  int64_t peek_size(double timeout) {
-    struct timespec deadline;
-    if (timeout > 0) make_deadline(timeout, &deadline);
-    struct timespec tick;
-    RobustLock lock(&hdr_->mu);
-    for (;;) {
-      if (hdr_->msg_count > 0) {
-        uint64_t cap = hdr_->capacity;
-        uint64_t head = hdr_->head;
-        uint32_t len32;
-        std::memcpy(&len32, data_ + head, 4);
-        if (len32 == kWrapMarker) {
-          hdr_->used -= cap - head;
-          hdr_->head = head = 0;
-          std::memcpy(&len32, data_ + head, 4);
-        }
-        uint32_t state =
-            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
-                ->load(std::memory_order_acquire);
-        if (state == kStCommitted) return (int64_t)len32;
-        make_deadline(0.1, &tick);
-        cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
-        if (timeout == 0) return -1;
-        continue;
-      }
-      if (hdr_->closed) throw std::runtime_error("ring closed");
-      if (timeout == 0) return -1;
-      int rc;
-      if (timeout < 0) {
-        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
-      } else {
-        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
-        if (rc == ETIMEDOUT) return -1;
-      }
-      (void)rc;
-    }
-  }
-
-  // Single-copy receive: copies the next committed message into buf.
-  // Returns the message length, -1 on timeout, or -(len) - 2 if buf is
-  // too small (message left in the ring).
+  // Single-copy receive.  Returns len, -1 on timeout, or -(len)-2 if the
+  // buffer is too small (message left in place).
   int64_t recv_into(char* buf, size_t buflen, double timeout) {
-    struct timespec deadline;
-    if (timeout > 0) make_deadline(timeout, &deadline);
-    struct timespec tick;
-    RobustLock lock(&hdr_->mu);
-    int stuck_polls = 0;
-    for (;;) {
-      if (hdr_->msg_count > 0) {
-        uint64_t cap = hdr_->capacity;
-        uint64_t head = hdr_->head;
-        uint32_t len32;
-        std::memcpy(&len32, data_ + head, 4);
-        if (len32 == kWrapMarker) {
-          hdr_->used -= cap - head;
-          hdr_->head = head = 0;
-          std::memcpy(&len32, data_ + head, 4);
-        }
-        uint32_t state =
-            reinterpret_cast<std::atomic<uint32_t>*>(data_ + head + 4)
-                ->load(std::memory_order_acquire);
-        if (state == kStCommitted) {
-          if ((size_t)len32 > buflen) return -((int64_t)len32) - 2;
-          std::memcpy(buf, data_ + head + kRecHdr, len32);
-          size_t need = record_bytes(len32);
-          hdr_->head = (head + need) % cap;
-          hdr_->used -= need;
-          hdr_->msg_count -= 1;
-          hdr_->total_out += 1;
-          pthread_cond_signal(&hdr_->not_full);
-          return (int64_t)len32;
-        }
-        if (stuck_polls > 20) {
-          uint32_t pid;
-          std::memcpy(&pid, data_ + head + 8, 4);
-          if (!process_alive(pid)) {
-            size_t need = record_bytes(len32);
-            hdr_->head = (head + need) % cap;
-            hdr_->used -= need;
-            hdr_->msg_count -= 1;
-            pthread_cond_signal(&hdr_->not_full);
-            stuck_polls = 0;
-            continue;
-          }
-          stuck_polls = 0;
-        }
-        make_deadline(0.1, &tick);
-        int rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &tick);
-        if (rc == ETIMEDOUT) {
-          ++stuck_polls;
-          if (timeout == 0) return -1;
-          if (timeout > 0) {
-            struct timespec now;
-            clock_gettime(CLOCK_MONOTONIC, &now);
-            if (now.tv_sec > deadline.tv_sec ||
-                (now.tv_sec == deadline.tv_sec &&
-                 now.tv_nsec >= deadline.tv_nsec))
-              return -1;
-          }
-        }
-        continue;
+    int64_t result = -1;
+    bool ok = recv_loop(timeout, [&](uint64_t head, uint32_t len32) {
+      if ((size_t)len32 > buflen) {
+        result = -((int64_t)len32) - 2;
+        return false;
       }
-      if (hdr_->closed) throw std::runtime_error("ring closed");
-      if (timeout == 0) return -1;
-      int rc;
-      if (timeout < 0) {
-        rc = cond_wait_robust(&hdr_->not_empty, &hdr_->mu);
-      } else {
-        rc = cond_timedwait_robust(&hdr_->not_empty, &hdr_->mu, &deadline);
-        if (rc == ETIMEDOUT) return -1;
-      }
-      (void)rc;
-    }
+      std::memcpy(buf, data_ + head + kRecHdr, len32);
+      consume_record(head, len32);
+      result = (int64_t)len32;
+      return true;
+    });
+    (void)ok;
+    return result;
+  }
+
+  // Size of the next committed message, or -1 on timeout.  No consume.
+  int64_t peek_size(double timeout) {
+    int64_t result = -1;
+    recv_loop(timeout, [&](uint64_t /*head*/, uint32_t len32) {
+      result = (int64_t)len32;
+      return true;
+    });
+    return result;
   }
 
   void close_ring() {
     if (!hdr_) return;
-    // Time-boxed: teardown must never wedge even if the mutex is stuck
-    // (e.g. an undiagnosed robust-mutex corner under SIGKILL chaos).
-    // The closed flag is a u32 checked inside every wait loop, so a
-    // lock-free store + broadcast still wakes and drains all waiters.
+    // Time-boxed lock: teardown must never wedge.  The closed flag is
+    // checked inside every wait loop, so a lock-free store + wake still
+    // drains all waiters even if the mutex is stuck.
     struct timespec deadline;
     clock_gettime(CLOCK_REALTIME, &deadline);
     deadline.tv_sec += 5;
@@ -525,10 +443,10 @@ class ShmRing {
       pthread_mutex_consistent(&hdr_->mu);
       rc = 0;
     }
-    hdr_->closed = 1;
-    pthread_cond_broadcast(&hdr_->not_empty);
-    pthread_cond_broadcast(&hdr_->not_full);
+    as_atomic(&hdr_->closed)->store(1, std::memory_order_release);
     if (rc == 0) pthread_mutex_unlock(&hdr_->mu);
+    bump_and_wake(&hdr_->fut_not_empty);
+    bump_and_wake(&hdr_->fut_not_full);
   }
 
   void unlink_ring() { shm_unlink(name_.c_str()); }
@@ -541,6 +459,7 @@ class ShmRing {
   const std::string& name() const { return name_; }
 
  private:
+  static constexpr double st2_wait_ = 1.0;
   std::string name_;
   bool owner_;
   Header* hdr_ = nullptr;

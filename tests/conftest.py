@@ -49,3 +49,12 @@ def leak_check():
     # sweep); they are cleaned above but not a test failure.
     new = {n for n in new if ".task.r." not in n}
     assert not new, "leaked shm segments: %r" % new
+
+
+# On-demand all-thread stack dumps for hang forensics:
+#   FAM_DEBUG_FAULTHANDLER=1 pytest ... ; kill -USR1 <pid>
+if os.environ.get("FAM_DEBUG_FAULTHANDLER"):
+    import faulthandler
+    import signal as _signal
+
+    faulthandler.register(_signal.SIGUSR1, all_threads=True)
