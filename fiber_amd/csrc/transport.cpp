@@ -68,10 +68,14 @@ struct Header {
   uint32_t magic;
   uint32_t ready;  // set to 1 once the creator finished initialization
   pthread_mutex_t mu;
-  // futex sequence words: bumped (under the mutex) whenever the
-  // corresponding condition may have become true, woken with FUTEX_WAKE
+  // futex sequence words: bumped whenever the corresponding condition
+  // may have become true, woken with FUTEX_WAKE (skipped when the
+  // waiter counter is zero — saves a syscall per message on the fast
+  // path)
   uint32_t fut_not_empty;
   uint32_t fut_not_full;
+  uint32_t waiters_not_empty;
+  uint32_t waiters_not_full;
   uint64_t capacity;   // bytes in data area
   uint64_t head;       // consumer offset
   uint64_t tail;       // producer offset
@@ -129,12 +133,23 @@ inline void futex_wake_all(uint32_t* word) {
   syscall(SYS_futex, word, FUTEX_WAKE, INT32_MAX, nullptr, nullptr, 0);
 }
 
-// Bump the sequence word (must hold the mutex so waiters' snapshot-then-
-// wait ordering is race-free) and wake everyone.
-inline void bump_and_wake(uint32_t* word) {
-  as_atomic(word)->fetch_add(1, std::memory_order_release);
-  futex_wake_all(word);
+// Bump the sequence word and wake everyone — the syscall is skipped
+// when no one waits.  A waiter that registers after our check re-reads
+// the word before sleeping (snapshot-first ordering), so it cannot
+// sleep through this bump.
+inline void bump_and_wake(uint32_t* word, uint32_t* waiters) {
+  as_atomic(word)->fetch_add(1, std::memory_order_seq_cst);
+  if (as_atomic(waiters)->load(std::memory_order_seq_cst) != 0)
+    futex_wake_all(word);
 }
+
+struct WaiterScope {
+  uint32_t* w;
+  explicit WaiterScope(uint32_t* waiters) : w(waiters) {
+    as_atomic(w)->fetch_add(1, std::memory_order_seq_cst);
+  }
+  ~WaiterScope() { as_atomic(w)->fetch_sub(1, std::memory_order_seq_cst); }
+};
 
 class RobustLock {
  public:
@@ -293,13 +308,18 @@ class ShmRing {
         remaining = deadline - monotonic_now();
         if (remaining <= 0) return false;
       }
-      futex_wait(&hdr_->fut_not_full, snap, remaining);
+      {
+        WaiterScope ws(&hdr_->waiters_not_full);
+        if (as_atomic(&hdr_->fut_not_full)
+                ->load(std::memory_order_seq_cst) == snap)
+          futex_wait(&hdr_->fut_not_full, snap, remaining);
+      }
     }
 
     if (len) std::memcpy(data_ + rec + kRecHdr, buf, len);
     as_atomic(reinterpret_cast<uint32_t*>(data_ + rec + 4))
         ->store(kStCommitted, std::memory_order_release);
-    bump_and_wake(&hdr_->fut_not_empty);
+    bump_and_wake(&hdr_->fut_not_empty, &hdr_->waiters_not_empty);
     return true;
   }
 
@@ -358,7 +378,7 @@ class ShmRing {
         if (st == 0) {
           if (op(head, len32)) {
             lock.unlock();
-            bump_and_wake(&hdr_->fut_not_full);
+            bump_and_wake(&hdr_->fut_not_full, &hdr_->waiters_not_full);
             return true;
           }
           return false;  // op declined (recv_into: buffer too small)
@@ -374,7 +394,8 @@ class ShmRing {
             if (!process_alive(pid)) {
               reclaim_dead_record(head, len32);
               lock.unlock();
-              bump_and_wake(&hdr_->fut_not_full);
+              bump_and_wake(&hdr_->fut_not_full,
+                            &hdr_->waiters_not_full);
               reserved_streak = 0;
               continue;
             }
@@ -390,7 +411,12 @@ class ShmRing {
         remaining = deadline - monotonic_now();
         if (remaining <= 0) return false;
       }
-      futex_wait(&hdr_->fut_not_empty, snap, remaining);
+      {
+        WaiterScope ws(&hdr_->waiters_not_empty);
+        if (as_atomic(&hdr_->fut_not_empty)
+                ->load(std::memory_order_seq_cst) == snap)
+          futex_wait(&hdr_->fut_not_empty, snap, remaining);
+      }
     }
   }
 
@@ -445,8 +471,11 @@ class ShmRing {
     }
     as_atomic(&hdr_->closed)->store(1, std::memory_order_release);
     if (rc == 0) pthread_mutex_unlock(&hdr_->mu);
-    bump_and_wake(&hdr_->fut_not_empty);
-    bump_and_wake(&hdr_->fut_not_full);
+    as_atomic(&hdr_->fut_not_empty)
+        ->fetch_add(1, std::memory_order_seq_cst);
+    as_atomic(&hdr_->fut_not_full)->fetch_add(1, std::memory_order_seq_cst);
+    futex_wake_all(&hdr_->fut_not_empty);  // unconditional on teardown
+    futex_wake_all(&hdr_->fut_not_full);
   }
 
   void unlink_ring() { shm_unlink(name_.c_str()); }
