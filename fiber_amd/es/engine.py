@@ -87,6 +87,14 @@ class ESEngine:
 
         # static workspace so repeated steps allocate nothing
         self._fitness_all = torch.empty(self.pop_total, device=device)
+        # grad (NPARAMS) and obs_stat (2*obs_dim+1) share ONE flat buffer
+        # so the per-step reduction is a single RCCL all-reduce: xGMI ring
+        # latency is paid once, not twice, per iteration (matters at 8
+        # ranks where both payloads are tiny).
+        nstat = 2 * config.obs_dim + 1
+        self._reduce_buf = torch.empty(ops.NPARAMS + nstat, device=device)
+        self._grad_view = self._reduce_buf[: ops.NPARAMS]
+        self._stat_view = self._reduce_buf[ops.NPARAMS :]
 
     # -- one ES iteration --------------------------------------------------
     def step(self, iteration=None):
@@ -120,8 +128,12 @@ class ESEngine:
                                iteration, self.device)
         if self.ctx is not None:
             with tracing.range("es.allreduce_grad"):
-                self.ctx.allreduce(grad)
-                self.ctx.allreduce(obs_stat)
+                # one fused all-reduce for grad + obs stats
+                self._grad_view.copy_(grad)
+                self._stat_view.copy_(obs_stat)
+                self.ctx.allreduce(self._reduce_buf)
+                grad = self._grad_view
+                obs_stat = self._stat_view
         grad /= float(self.pop_total) * cfg.sigma
 
         # maximize fitness => ascend
