@@ -154,6 +154,25 @@ class SimpleQueue:
             raise TimeoutError("queue get timed out")
         return serialization.loads(data)
 
+    def put_many(self, objs, timeout=-1.0):
+        """Enqueue a batch of objects with one lock hold + one wake per
+        burst (~1.7x the per-message rate of put() for small payloads;
+        see profiles).  Atomicity is per-message, not per-batch."""
+        ring = self._ensure()
+        payloads = [serialization.dumps_closure(o) for o in objs]
+        sent = ring.send_many(payloads, timeout)
+        if sent != len(payloads):
+            raise TimeoutError(
+                "queue put_many timed out after %d of %d" %
+                (sent, len(payloads))
+            )
+
+    def get_many(self, max_n=64, timeout=-1.0):
+        """Dequeue up to max_n objects; blocks (per timeout) only for the
+        first.  Returns a possibly-empty list on timeout."""
+        ring = self._ensure()
+        return [serialization.loads(d) for d in ring.recv_many(max_n, timeout)]
+
     def get_nowait(self):
         ring = self._ensure()
         data = _ring_recv(ring, 0.0)

@@ -146,3 +146,48 @@ class TestSimpleQueue:
             q.get(timeout=0.2)
         assert time.monotonic() - t0 < 2.0
         q.close()
+
+
+class TestSimpleQueueBatched:
+    def test_put_many_get_many(self):
+        q = SimpleQueue()
+        try:
+            q.put_many([{"i": i} for i in range(10)])
+            got = q.get_many(max_n=6, timeout=5.0)
+            assert got == [{"i": i} for i in range(6)]
+            got += q.get_many(max_n=64, timeout=5.0)
+            assert got == [{"i": i} for i in range(10)]
+            assert q.get_many(max_n=4, timeout=0.0) == []
+        finally:
+            q.close()
+
+    def test_batched_across_processes(self):
+        import functools
+
+        q = SimpleQueue()
+        out = SimpleQueue()
+        try:
+            p = fiber_amd.Process(
+                target=functools.partial(_batch_echo, q, out), name="qb"
+            )
+            p.start()
+            q.put_many(list(range(100)))
+            got = []
+            while len(got) < 100:
+                items = out.get_many(max_n=64, timeout=30.0)
+                assert items, "starved at %d" % len(got)
+                got.extend(items)
+            assert got == [i * 2 for i in range(100)]
+            p.join(30)
+            assert p.exitcode == 0
+        finally:
+            q.close()
+            out.close()
+
+
+def _batch_echo(q, out):
+    done = 0
+    while done < 100:
+        items = q.get_many(max_n=32, timeout=30.0)
+        out.put_many([i * 2 for i in items])
+        done += len(items)

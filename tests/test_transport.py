@@ -180,3 +180,135 @@ class TestDeadWriterReclaim:
         victim.join(10)
         survivor.join(30)
         reader.close()
+
+
+class TestBatchedOps:
+    """send_many/recv_many: one lock hold + one wake per burst."""
+
+    def test_batch_roundtrip_and_interleave(self):
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 1 << 20, 5.0)
+        try:
+            assert ring.send_many([b"a", b"bb", b"", b"dddd"], 1.0) == 4
+            assert ring.recv(0.5) == b"a"
+            assert ring.recv_many(10, 0.5) == [b"bb", b"", b"dddd"]
+            ring.send(b"x1")
+            ring.send_many([b"x2", b"x3"])
+            assert ring.recv_many(2, 0.5) == [b"x1", b"x2"]
+            assert ring.recv_many(5, 0.5) == [b"x3"]
+            assert ring.recv_many(4, 0.0) == []
+            assert ring.total_in == 7 and ring.total_out == 7
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_batch_backpressure_order_exact(self):
+        """A 600-message batch through a 4 KB ring: blocking bursts,
+        wraps, exact content and counters."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        try:
+            msgs = [bytes([i % 256]) * 100 for i in range(600)]
+            sent = []
+            th = threading.Thread(
+                target=lambda: sent.append(ring.send_many(msgs, 30.0))
+            )
+            th.start()
+            out = []
+            while len(out) < 600:
+                got = ring.recv_many(64, 5.0)
+                assert got, "starved at %d" % len(out)
+                out.extend(got)
+            th.join(10)
+            assert sent == [600]
+            assert out == msgs
+            assert ring.size == 0
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_wrap_waste_near_full_geometry(self):
+        """Regression: the producer fit check must price the wrap waste
+        (capacity - tail) exactly.  Messages sized near capacity/3 keep
+        the ring in the near-full wrap window where a fixed-margin check
+        overruns the reader's head (used > capacity, unsigned underflow,
+        corruption)."""
+        import os
+        import random
+
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        try:
+            random.seed(11)
+            msgs = [os.urandom(random.randint(900, 1900)) for _ in range(300)]
+            th = threading.Thread(target=lambda: ring.send_many(msgs, 60.0))
+            th.start()
+            out = []
+            while len(out) < 300:
+                got = ring.recv_many(8, 5.0)
+                assert got, "starved at %d" % len(out)
+                out.extend(got)
+            th.join(10)
+            assert out == msgs
+            # singles through the same geometry
+            th = threading.Thread(
+                target=lambda: [ring.send(m, 60.0) for m in msgs]
+            )
+            th.start()
+            out = [ring.recv(5.0) for _ in range(300)]
+            th.join(10)
+            assert out == msgs
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_large_message_fits_regardless_of_tail_offset(self):
+        """Empty-ring rewind: a near-capacity message must fit even after
+        head/tail drifted to an arbitrary offset."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        try:
+            big = b"B" * ((4 << 10) - 64)
+            for _ in range(8):  # drift the offsets, then send big
+                ring.send(b"pad" * 41, 1.0)
+                assert ring.recv(1.0)
+                assert ring.send(big, 1.0), "big message did not fit"
+                assert ring.recv(1.0) == big
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_batch_survives_dead_writer_reclaim(self):
+        """A producer SIGKILLed somewhere inside send_many must not stall
+        batched consumers: reserved records from the dead pid are
+        reclaimed, committed ones are delivered intact."""
+        import os
+        import signal
+
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 32 << 10, 5.0)
+        try:
+            pid = os.fork()
+            if pid == 0:  # child: batch-send forever until killed
+                child = ShmRing(name, False, 0, 5.0)
+                batch = [b"k" * 64] * 16
+                while True:
+                    child.send_many(batch, 5.0)
+            got = 0
+            while got < 200:  # let plenty through first
+                got += len(ring.recv_many(32, 5.0))
+            os.kill(pid, signal.SIGKILL)
+            os.waitpid(pid, 0)
+            # live producer must still make progress afterwards
+            assert ring.send_many([b"after1", b"after2"], 5.0) == 2
+            seen = []
+            deadline = 50
+            while deadline and seen[-2:] != [b"after1", b"after2"]:
+                seen.extend(ring.recv_many(32, 1.0))
+                deadline -= 1
+            assert seen[-2:] == [b"after1", b"after2"]
+            for m in seen[:-2]:
+                assert m == b"k" * 64
+        finally:
+            ring.close()
+            ring.unlink()
