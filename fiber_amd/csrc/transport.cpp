@@ -457,6 +457,7 @@ class ShmRing {
     const double deadline =
         timeout > 0 ? monotonic_now() + timeout : 0.0;
     int reserved_streak = 0;
+    int st = 1;
     for (;;) {
       uint32_t snap = as_atomic(&hdr_->fut_not_empty)
                           ->load(std::memory_order_acquire);
@@ -464,7 +465,7 @@ class ShmRing {
         RobustLock lock(&hdr_->mu);
         uint64_t head;
         uint32_t len32;
-        int st = resolve_head(&head, &len32);
+        st = resolve_head(&head, &len32);
         if (st == 0) {
           if (op(head, len32)) {
             lock.unlock();
@@ -496,10 +497,17 @@ class ShmRing {
         }
         if (timeout == 0) return false;
       }
-      double remaining = st2_wait_;
+      // A reserved head means the writer is mid-copy: a LIVE writer's
+      // commit bumps the futex and wakes us immediately, so capping this
+      // wait at kReservedWait only slows the dead-writer path — it lets
+      // reserved_streak reach the liveness check even when the caller
+      // uses short per-call timeouts (< the 1 s wait chunk).
+      double cap_wait = (st == 2) ? kReservedWait : 1.0;
+      double remaining = cap_wait;
       if (timeout > 0) {
         remaining = deadline - monotonic_now();
         if (remaining <= 0) return false;
+        if (remaining > cap_wait) remaining = cap_wait;
       }
       {
         WaiterScope ws(&hdr_->waiters_not_empty);
@@ -520,6 +528,7 @@ class ShmRing {
     if (max_n > kMaxBatch) max_n = kMaxBatch;
     const double deadline = timeout > 0 ? monotonic_now() + timeout : 0.0;
     int reserved_streak = 0;
+    int st = 1;
     for (;;) {
       uint32_t snap = as_atomic(&hdr_->fut_not_empty)
                           ->load(std::memory_order_acquire);
@@ -527,7 +536,7 @@ class ShmRing {
         RobustLock lock(&hdr_->mu);
         uint64_t head;
         uint32_t len32;
-        int st = resolve_head(&head, &len32);
+        st = resolve_head(&head, &len32);
         if (st == 0) {
           do {
             out->emplace_back(data_ + head + kRecHdr, len32);
@@ -560,10 +569,12 @@ class ShmRing {
         }
         if (timeout == 0) return 0;
       }
-      double remaining = 1.0;
+      double cap_wait = (st == 2) ? kReservedWait : 1.0;  // see recv_loop
+      double remaining = cap_wait;
       if (timeout > 0) {
         remaining = deadline - monotonic_now();
         if (remaining <= 0) return 0;
+        if (remaining > cap_wait) remaining = cap_wait;
       }
       {
         WaiterScope ws(&hdr_->waiters_not_empty);
@@ -642,7 +653,7 @@ class ShmRing {
   const std::string& name() const { return name_; }
 
  private:
-  static constexpr double st2_wait_ = 1.0;
+  static constexpr double kReservedWait = 0.1;
   std::string name_;
   bool owner_;
   Header* hdr_ = nullptr;
