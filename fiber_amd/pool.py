@@ -647,31 +647,31 @@ class ZPool:
         self._task_sock.send(serialization.dumps(task), timeout=-1.0)
 
     def _result_loop(self):
-        from .transport import ring_recv_view
-
+        ring = self._result_sock._rings["main"]
         while True:
             try:
-                payload = ring_recv_view(
-                    self._result_sock._rings["main"], 0.2
-                )
+                # batched drain: one lock hold + one wake per burst of
+                # completed chunks (matters at small chunksizes)
+                payloads = ring.recv_many(32, 0.2)
             except RuntimeError:
                 return
             except Exception:
                 return
             if self._state == "terminated":
                 return
-            if payload is None:
-                continue
-            seq, base, values, failure, ident = serialization.loads(payload)
-            self._recv += 1
-            self._ack(ident, seq, base)
-            self._inventory.put(seq, base, values, failure)
-            watcher = self._callback_watch.get(seq)
-            if watcher is not None:
-                done, error, results = self._inventory.peek(seq)
-                if done:
-                    self._callback_watch.pop(seq, None)
-                    watcher._fire_callbacks(error, results)
+            for payload in payloads:
+                seq, base, values, failure, ident = (
+                    serialization.loads(payload)
+                )
+                self._recv += 1
+                self._ack(ident, seq, base)
+                self._inventory.put(seq, base, values, failure)
+                watcher = self._callback_watch.get(seq)
+                if watcher is not None:
+                    done, error, results = self._inventory.peek(seq)
+                    if done:
+                        self._callback_watch.pop(seq, None)
+                        watcher._fire_callbacks(error, results)
 
     def _ack(self, ident, seq, base):
         pass  # resilient subclass clears pending table
