@@ -61,3 +61,32 @@ class TestHipIpcTensors:
 def _tensor_sum(t):
     assert t.is_cuda
     return float(t.sum().item())
+
+
+def _batch_consumer(q, out):
+    got = []
+    while len(got) < 8:
+        got.extend(q.get_many(max_n=8, timeout=60))
+    assert all(t.is_cuda for t in got)
+    out.put([float(t.sum().item()) for t in got])
+
+
+@requires_gpu
+class TestBatchedIpcTensors:
+    def test_put_many_cuda_tensors(self):
+        """Batched queue ops compose with the HIP-IPC tensor reducers:
+        each batched message still carries only an IPC handle."""
+        q, out = SimpleQueue(), SimpleQueue()
+        tensors = [
+            torch.full((256,), float(i), device="cuda") for i in range(8)
+        ]
+        torch.cuda.synchronize()
+        p = fiber_amd.Process(target=_batch_consumer, args=(q, out))
+        p.start()
+        q.put_many(tensors)
+        sums = out.get(timeout=120)
+        assert sums == [256.0 * i for i in range(8)]
+        p.join(60)
+        assert p.exitcode == 0
+        q.close()
+        out.close()
