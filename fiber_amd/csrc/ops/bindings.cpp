@@ -34,9 +34,10 @@ extern "C" __global__ void es_perturb(const float*, int, int, float,
                                       unsigned char*);
 extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
                                          float*, float*);
-extern "C" __global__ void conv_obsgen(const float*, const float*, uint32_t,
-                                       const uint32_t*, uint32_t,
-                                       unsigned char*);
+extern "C" __global__ void conv_noisegen(uint32_t, const uint32_t*,
+                                         uint32_t, float*);
+extern "C" __global__ void conv_obsgen(const float*, const float*,
+                                       const float*, unsigned char*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
                                        const unsigned char*,
                                        const unsigned char*, int,
@@ -144,13 +145,21 @@ static void launch_conv_env_init(uint32_t seed, uintptr_t iterp,
   check(hipGetLastError(), "conv_env_init launch");
 }
 
+static void launch_conv_noisegen(uint32_t seed, uintptr_t iterp, uint32_t t,
+                                 int nenv, uintptr_t znoise,
+                                 uintptr_t stream) {
+  hipLaunchKernelGGL(conv_noisegen, dim3(nenv), dim3(256), 0,
+                     (hipStream_t)stream, seed, (const uint32_t*)iterp, t,
+                     (float*)znoise);
+  check(hipGetLastError(), "conv_noisegen launch");
+}
+
 static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
-                               uint32_t seed, uintptr_t iterp, uint32_t t,
-                               int nenv_total, uintptr_t obs,
-                               uintptr_t stream) {
+                               uintptr_t znoise, int nenv_total,
+                               uintptr_t obs, uintptr_t stream) {
   hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
                      (hipStream_t)stream, (const float*)state,
-                     (const float*)gtab, seed, (const uint32_t*)iterp, t,
+                     (const float*)gtab, (const float*)znoise,
                      (unsigned char*)obs);
   check(hipGetLastError(), "conv_obsgen launch");
 }
@@ -224,9 +233,12 @@ PYBIND11_MODULE(_ops, m) {
   m.def("conv_env_init", &launch_conv_env_init, py::arg("seed"),
         py::arg("iterp"), py::arg("nmembers"), py::arg("state"),
         py::arg("racc"), py::arg("stream"));
+  m.def("conv_noisegen", &launch_conv_noisegen, py::arg("seed"),
+        py::arg("iterp"), py::arg("t"), py::arg("nenv"),
+        py::arg("znoise"), py::arg("stream"));
   m.def("conv_obsgen", &launch_conv_obsgen, py::arg("state"),
-        py::arg("gtab"), py::arg("seed"), py::arg("iterp"), py::arg("t"),
-        py::arg("nenv_total"), py::arg("obs"), py::arg("stream"));
+        py::arg("gtab"), py::arg("znoise"), py::arg("nenv_total"),
+        py::arg("obs"), py::arg("stream"));
   m.def("conv_forward", &launch_conv_forward, py::arg("wpert"),
         py::arg("w3_fp8"), py::arg("w1_fp8"), py::arg("obs"),
         py::arg("act1"), py::arg("act2"), py::arg("act3"),

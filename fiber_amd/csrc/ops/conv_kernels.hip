@@ -131,29 +131,53 @@ extern "C" __global__ void conv_env_init(uint32_t seed,
 }
 
 // ---------------------------------------------------------------------------
-// conv_obsgen: obs[b][e][y][x][ic] = 0.3*noise(e,t,pos) +
-//                                    state[b][e][ic] * gtab[y][x]
-// one workgroup per (member, env); channel-last writes, 4 bf16 per pos.
+// conv_noisegen + conv_obsgen: obs[b][e][y][x][ic] =
+//     fp8( 0.52*noise(e,t,pos,ic) + state[b][e][ic] * gtab[y][x] )
+//
+// The noise is keyed by (env, position, t, iter) ONLY — common random
+// numbers across the whole ES population, so every member of a chunk
+// sees the same scenery draw.  Exploit that: conv_noisegen runs the
+// philox-7 chains ONCE per (chunk, t) into a 16-env fp32 buffer
+// (1/nmembers of the RNG work obsgen used to redo per member), and
+// conv_obsgen becomes a pure bandwidth kernel that combines the shared
+// noise with each member's state.  The arithmetic expression and
+// rounding points are unchanged, so results are bit-identical to the
+// fused version (and to conv_rollout_reference).
 // ---------------------------------------------------------------------------
+extern "C" __global__ void conv_noisegen(uint32_t seed,
+                                         const uint32_t* __restrict__ iterp,
+                                         uint32_t t,
+                                         float* __restrict__ znoise) {
+  const uint32_t iter = *iterp;
+  const int e = blockIdx.x;  // one workgroup per env slot
+  float* out = znoise + (size_t)e * (IMG * IMG * CIN);
+  // 4 positions per iteration -> 4 independent philox chains in flight
+  // (the mad_u64 round chain is serial; cross-position ILP fills the
+  // pipe).  IMG*IMG = 7056 is a multiple of 4 -> exact coverage.
+  for (int p = threadIdx.x * 4; p + 3 < IMG * IMG; p += blockDim.x * 4) {
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      float z[4];
+      fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)(p + q), FAM_TAG_OBS,
+                   t, z);
+      *reinterpret_cast<float4*>(&out[(p + q) * CIN]) =
+          make_float4(z[0], z[1], z[2], z[3]);
+    }
+  }
+}
+
+// one workgroup per (member, env); channel-last fp8 writes, 16 B stores.
 extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
                                        const float* __restrict__ gtab,
-                                       uint32_t seed,
-                                       const uint32_t* __restrict__ iterp,
-                                       uint32_t t,
+                                       const float* __restrict__ znoise,
                                        unsigned char* __restrict__ obs) {
-  const uint32_t iter = *iterp;
   const int be = blockIdx.x;  // member*CENV + env
   const int e = be % CENV;
+  const float* zn = znoise + (size_t)e * (IMG * IMG * CIN);
   unsigned char* out = obs + (size_t)be * (IMG * IMG * CIN);
   float s[SDIM];
 #pragma unroll
   for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
-  // each position needs CIN=4 values (one philox draw); FOUR adjacent
-  // positions per thread iteration -> 4 independent philox chains in
-  // flight (the 10-round mad_u64 chain is serial, so ILP across
-  // positions is what fills the pipe) and one 16 B store.
-  // IMG*IMG = 7056 is a multiple of 4, so the strided 4-position loop
-  // covers every position exactly once (thread t owns p ≡ 4t mod 4096).
   for (int p = threadIdx.x * 4; p + 3 < IMG * IMG; p += blockDim.x * 4) {
     union {
       unsigned char b[16];
@@ -161,13 +185,12 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
     } pk;
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
-      float z[4];
-      fam_uniform4(seed, iter, (uint32_t)e, (uint32_t)(p + q), FAM_TAG_OBS,
-                   t, z);
+      const float4 z = *reinterpret_cast<const float4*>(&zn[(p + q) * CIN]);
       const float g = gtab[p + q];
-#pragma unroll
-      for (int c = 0; c < CIN; ++c)
-        pk.b[q * 4 + c] = __hip_fp8_e4m3(0.52f * z[c] + s[c] * g).__x;
+      pk.b[q * 4 + 0] = __hip_fp8_e4m3(0.52f * z.x + s[0] * g).__x;
+      pk.b[q * 4 + 1] = __hip_fp8_e4m3(0.52f * z.y + s[1] * g).__x;
+      pk.b[q * 4 + 2] = __hip_fp8_e4m3(0.52f * z.z + s[2] * g).__x;
+      pk.b[q * 4 + 3] = __hip_fp8_e4m3(0.52f * z.w + s[3] * g).__x;
     }
     *reinterpret_cast<fam_uint4*>(&out[p * CIN]) =
         *reinterpret_cast<fam_uint4*>(pk.w);

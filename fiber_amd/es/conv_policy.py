@@ -138,6 +138,14 @@ class ConvESEngine:
         self._half_streams = [
             torch.cuda.Stream(device=device) for _ in range(nhalves)
         ]
+        # shared obs-noise staging, one fp32 field per stream chunk: the
+        # noise is keyed by (env, pos, t) only (common random numbers
+        # across the population), so conv_noisegen fills this once per
+        # (chunk, t) and conv_obsgen re-reads it for every member —
+        # 1/(pop/chunk) of the philox work the fused kernel redid.
+        # Chunks run at skewed timesteps, hence one buffer each.
+        self.znoise = torch.empty(nhalves, E * 84 * 84 * 4,
+                                  dtype=torch.float32, device=device)
 
     def _stream(self):
         return torch.cuda.current_stream().cuda_stream
@@ -169,9 +177,11 @@ class ConvESEngine:
                      cfg.sigma, cfg.seed, iterp, member_offset, pop,
                      wpert, w3_fp8, w1_fp8, stream)
         o.conv_env_init(cfg.seed, iterp, pop, state, racc, stream)
+        znoise = self.znoise[half].data_ptr()
         for t in range(cfg.horizon):
-            o.conv_obsgen(state, self.gtab.data_ptr(), cfg.seed, iterp, t,
-                          pop * E, obs, stream)
+            o.conv_noisegen(cfg.seed, iterp, t, E, znoise, stream)
+            o.conv_obsgen(state, self.gtab.data_ptr(), znoise, pop * E,
+                          obs, stream)
             o.conv_forward(wpert, w3_fp8, w1_fp8, obs, act1, act2, act3,
                            pop, stream)
             o.conv_head_env(wpert, act3, pop, self.env_A.data_ptr(),
