@@ -35,9 +35,10 @@ extern "C" __global__ void es_perturb(const float*, int, int, float,
 extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
                                          float*, float*);
 extern "C" __global__ void conv_noisegen(uint32_t, const uint32_t*,
-                                         uint32_t, float*);
+                                         uint32_t, unsigned char*);
 extern "C" __global__ void conv_obsgen(const float*, const float*,
-                                       const float*, unsigned char*);
+                                       const unsigned char*,
+                                       unsigned char*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
                                        const unsigned char*,
                                        const unsigned char*, int,
@@ -148,9 +149,11 @@ static void launch_conv_env_init(uint32_t seed, uintptr_t iterp,
 static void launch_conv_noisegen(uint32_t seed, uintptr_t iterp, uint32_t t,
                                  int nenv, uintptr_t znoise,
                                  uintptr_t stream) {
-  hipLaunchKernelGGL(conv_noisegen, dim3(nenv), dim3(256), 0,
+  // flattened (env, position-quad) grid; 7056/4 quads per env
+  const int total = nenv * (84 * 84 / 4);
+  hipLaunchKernelGGL(conv_noisegen, dim3((total + 255) / 256), dim3(256), 0,
                      (hipStream_t)stream, seed, (const uint32_t*)iterp, t,
-                     (float*)znoise);
+                     (unsigned char*)znoise);
   check(hipGetLastError(), "conv_noisegen launch");
 }
 
@@ -159,7 +162,7 @@ static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
                                uintptr_t obs, uintptr_t stream) {
   hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
                      (hipStream_t)stream, (const float*)state,
-                     (const float*)gtab, (const float*)znoise,
+                     (const float*)gtab, (const unsigned char*)znoise,
                      (unsigned char*)obs);
   check(hipGetLastError(), "conv_obsgen launch");
 }
