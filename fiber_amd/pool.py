@@ -625,7 +625,12 @@ class ZPool:
 
     # -- data plane --------------------------------------------------------
     def _dispatch_loop(self):
-        """Drain the local task queue into the task ring."""
+        """Drain the local task queue into the task ring.  Bursts of
+        queued chunks go out through one batched ring op (one lock hold
+        + one wake); the resilient subclass overrides _send_tasks to
+        keep its per-request credit scheduling."""
+        import queue as _stdq
+
         while True:
             task = self._taskq.get()
             if task is None:
@@ -637,11 +642,28 @@ class ZPool:
                 time.sleep(0.005)
             if self._state == "terminated":
                 continue
+            batch = [task]
+            sentinel = False
+            while len(batch) < 32:
+                try:
+                    nxt = self._taskq.get_nowait()
+                except _stdq.Empty:
+                    break
+                if nxt is None:
+                    sentinel = True
+                    break
+                batch.append(nxt)
             try:
-                self._send_task(task)
+                self._send_tasks(batch)
             except RuntimeError:
                 return  # ring closed
-            self._sent += 1
+            self._sent += len(batch)
+            if sentinel:
+                return
+
+    def _send_tasks(self, tasks):
+        payloads = [serialization.dumps(t) for t in tasks]
+        self._task_sock._rings["main"].send_many(payloads, -1.0)
 
     def _send_task(self, task):
         self._task_sock.send(serialization.dumps(task), timeout=-1.0)
@@ -889,6 +911,13 @@ class ResilientZPool(ZPool):
         with self._worker_lock:
             proc = self._workers.get(primary)
         return proc is not None and proc.exitcode is None
+
+    def _send_tasks(self, tasks):
+        # REQ/REP credit scheduling is inherently per-request: each chunk
+        # is handed to a specific live requester and recorded in its
+        # pending table, so batching degenerates to a loop.
+        for task in tasks:
+            self._send_task(task)
 
     def _send_task(self, task):
         # Serve the next *live* worker request, record attribution, reply.
