@@ -36,7 +36,8 @@ extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
                                          float*, float*);
 extern "C" __global__ void conv_noisegen(uint32_t, const uint32_t*,
                                          uint32_t, unsigned char*);
-extern "C" __global__ void conv_obsgen(const float*, const float*,
+extern "C" __global__ void conv_obsgen(const float*,
+                                       const __hip_bfloat16*,
                                        const unsigned char*,
                                        unsigned char*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
@@ -162,8 +163,8 @@ static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
                                uintptr_t obs, uintptr_t stream) {
   hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
                      (hipStream_t)stream, (const float*)state,
-                     (const float*)gtab, (const unsigned char*)znoise,
-                     (unsigned char*)obs);
+                     (const __hip_bfloat16*)gtab,
+                     (const unsigned char*)znoise, (unsigned char*)obs);
   check(hipGetLastError(), "conv_obsgen launch");
 }
 

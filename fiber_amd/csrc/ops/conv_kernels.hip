@@ -151,13 +151,6 @@ extern "C" __global__ void conv_env_init(uint32_t seed,
 // independent philox chains in flight (the mad_u64 round chain is
 // serial; cross-position ILP fills the pipe).  IMG*IMG = 7056 = 1764
 // quads per env.
-// Work is FLATTENED across (env, position-quad) so the kernel's duration
-// is one 4-draw philox chain, not a 1764-quad serial loop: this kernel
-// sits on the critical path ahead of every obsgen, so its latency (not
-// its total work) is what matters.  4 positions per thread -> 4
-// independent philox chains in flight (the mad_u64 round chain is
-// serial; cross-position ILP fills the pipe).  IMG*IMG = 7056 = 1764
-// quads per env.
 //
 // The scaled noise 0.52*z is staged as e4m3 (one byte per channel): the
 // obs themselves are e4m3-quantized right after the state term is
@@ -197,7 +190,8 @@ extern "C" __global__ void conv_noisegen(uint32_t seed,
 // Pure bandwidth: reads the (L2-resident) shared e4m3 noise field and
 // combines it with the member's state.
 extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
-                                       const float* __restrict__ gtab,
+                                       const __hip_bfloat16* __restrict__
+                                           gtab,
                                        const unsigned char* __restrict__
                                            znoise,
                                        unsigned char* __restrict__ obs) {
@@ -209,24 +203,29 @@ extern "C" __global__ void conv_obsgen(const float* __restrict__ state,
 #pragma unroll
   for (int d = 0; d < SDIM; ++d) s[d] = state[be * SDIM + d];
   for (int p = threadIdx.x * 4; p + 3 < IMG * IMG; p += blockDim.x * 4) {
-    union {
-      unsigned char b[16];
-      uint32_t w[4];
-    } pk;
+    fam_uint4 pk;
+    uint32_t* pw = &pk.x;
     const fam_uint4 zraw = *reinterpret_cast<const fam_uint4*>(&zn[p * CIN]);
-    const unsigned char* zb = reinterpret_cast<const unsigned char*>(&zraw);
+    const uint32_t* zw = &zraw.x;
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
-      const float g = gtab[p + q];
-#pragma unroll
-      for (int c = 0; c < CIN; ++c) {
-        __hip_fp8_e4m3 n;
-        n.__x = zb[q * 4 + c];
-        pk.b[q * 4 + c] = __hip_fp8_e4m3(float(n) + s[c] * g).__x;
-      }
+      // gtab rides bf16 (obsgen is at its HBM byte roofline; the
+      // pattern is smooth, bf16's 8-bit mantissa is plenty — the
+      // reference mirror applies the same rounding)
+      const float g = __bfloat162float(gtab[p + q]);
+      // raw v_cvt builtins: obs values are bounded (|obs| < ~4, e4m3
+      // max 448, never NaN), so the __hip_fp8_e4m3 constructor's
+      // software saturation chain (med3/cmp_class/cndmask per element)
+      // is dead weight — the hardware converter rounds RNE identically
+      // in range, keeping bytes bit-identical to the reference mirror.
+      const float r0 = __builtin_amdgcn_cvt_f32_fp8(zw[q], 0) + s[0] * g;
+      const float r1 = __builtin_amdgcn_cvt_f32_fp8(zw[q], 1) + s[1] * g;
+      const float r2 = __builtin_amdgcn_cvt_f32_fp8(zw[q], 2) + s[2] * g;
+      const float r3 = __builtin_amdgcn_cvt_f32_fp8(zw[q], 3) + s[3] * g;
+      uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(r0, r1, 0u, false);
+      pw[q] = __builtin_amdgcn_cvt_pk_fp8_f32(r2, r3, w, true);
     }
-    *reinterpret_cast<fam_uint4*>(&out[p * CIN]) =
-        *reinterpret_cast<fam_uint4*>(pk.w);
+    *reinterpret_cast<fam_uint4*>(&out[p * CIN]) = pk;
   }
 }
 

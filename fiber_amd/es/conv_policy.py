@@ -103,6 +103,8 @@ class ConvESEngine:
         self.t_step = 0
         self.env_A, self.env_B = make_env_params(config.seed, device)
         self.gtab = make_gtab(config.seed, device)
+        # obsgen reads the pattern in bf16 (it is HBM-byte-bound)
+        self.gtab_bf = self.gtab.to(torch.bfloat16).contiguous()
 
         bf = torch.bfloat16
         self.wpert = torch.empty(pop, o.NP_CONV_PAD, dtype=bf, device=device)
@@ -180,8 +182,8 @@ class ConvESEngine:
         znoise = self.znoise[half].data_ptr()
         for t in range(cfg.horizon):
             o.conv_noisegen(cfg.seed, iterp, t, E, znoise, stream)
-            o.conv_obsgen(state, self.gtab.data_ptr(), znoise, pop * E,
-                          obs, stream)
+            o.conv_obsgen(state, self.gtab_bf.data_ptr(), znoise,
+                          pop * E, obs, stream)
             o.conv_forward(wpert, w3_fp8, w1_fp8, obs, act1, act2, act3,
                            pop, stream)
             o.conv_head_env(wpert, act3, pop, self.env_A.data_ptr(),
@@ -366,7 +368,8 @@ def conv_rollout_reference(theta, sigma, seed, iteration, horizon, members,
             z = obs_noise(t)  # [E][7056][4]
             # the kernel stages 0.52*z as e4m3 (conv_noisegen), then
             # adds the state term and quantizes again
-            obs = fp8r(0.52 * z) + s[:, None, :] * gtab.reshape(1, -1, 1)
+            gb = bf(gtab).reshape(1, -1, 1)  # kernel reads gtab in bf16
+            obs = fp8r(0.52 * z) + s[:, None, :] * gb
             obs = fp8r(obs)
             x = obs.reshape(E, 84, 84, 4).permute(0, 3, 1, 2)
             h1 = bf(torch.tanh(
