@@ -403,28 +403,60 @@ conv_fc(const __hip_bfloat16* __restrict__ wpert,
   const int mt = blockIdx.y * 4 + wave;
   const int arow = mt * 16 + (lane & 15);
   const unsigned char* ap = &w3[(size_t)arow * NFLAT + kgrp * 8];
-  const unsigned char* bp = &in[(size_t)env * NFLAT + kgrp * 8];
-  // NOTE: double-buffered loads and dual accumulators were both tried
-  // here and measured SLOWER on hardware (369.5K / 368.3K vs 373.4K
-  // whole-bench rollouts/s) — at 4 wg/CU the load bursts are already
-  // hidden by wave parallelism and the extra registers only hurt.
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int kk = 0; kk < 80; kk += 8) {
-    fp8x8 a[8], b[8];
-#pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      a[u] = *reinterpret_cast<const fp8x8*>(ap + (kk + u) * 32);
-      b[u] = *reinterpret_cast<const fp8x8*>(bp + (kk + u) * 32);
+  // NOTE: register double-buffering and dual accumulators were tried
+  // and measured SLOWER (369.5K / 368.3K vs 373.4K whole-bench) — at
+  // 4 wg/CU the load bursts are wave-parallel-hidden.  What DOES pay
+  // is LDS-staging the B operand (act2): all 16 waves of a member used
+  // to re-pull the same 41.5 KB from L2/HBM while the W stream was
+  // evicting it.  B is staged in 27-tile k-chunks (16 envs x 864 k,
+  // padded to 872 B rows so the 16 env-columns of a fragment read hit
+  // 16 distinct LDS banks), double-buffered: 28 KB LDS keeps 4 wg/CU.
+  constexpr int kChunkK = 864;          // 27 of 81 k-tiles
+  constexpr int kRowPad = 872;          // bank-staggered env row stride
+  __shared__ alignas(16) unsigned char bsh[2][CENV * kRowPad];
+  auto stage = [&](int buf, int c0) {
+    // 16 envs x 864 B = 1728 8-byte words, cooperatively
+    for (int w8 = tid; w8 < CENV * (kChunkK / 8); w8 += 256) {
+      const int e8 = w8 / (kChunkK / 8);
+      const int j8 = w8 % (kChunkK / 8);
+      *reinterpret_cast<fp8x8*>(&bsh[buf][e8 * kRowPad + j8 * 8]) =
+          *reinterpret_cast<const fp8x8*>(
+              &in[(size_t)e8 * NFLAT + c0 + j8 * 8]);
     }
+  };
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  stage(0, 0);
+  __syncthreads();
+  for (int c = 0; c < 3; ++c) {
+    if (c + 1 < 3) stage((c + 1) & 1, (c + 1) * kChunkK);
+    const unsigned char* bl = &bsh[c & 1][env * kRowPad + kgrp * 8];
+    const int kt0 = c * 27;
+#pragma unroll 1
+    for (int kk = 0; kk < 24; kk += 8) {
+      fp8x8 a[8], b[8];
 #pragma unroll
-    for (int u = 0; u < 8; ++u)
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a[u], b[u], acc, 0,
-                                                       0, 0);
-  }
-  {  // tail (81st K-tile)
-    const fp8x8 a = *reinterpret_cast<const fp8x8*>(ap + 80 * 32);
-    const fp8x8 b = *reinterpret_cast<const fp8x8*>(bp + 80 * 32);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc, 0, 0, 0);
+      for (int u = 0; u < 8; ++u) {
+        a[u] = *reinterpret_cast<const fp8x8*>(ap + (kt0 + kk + u) * 32);
+        b[u] = *reinterpret_cast<const fp8x8*>(bl + (kk + u) * 32);
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a[u], b[u], acc,
+                                                         0, 0, 0);
+    }
+    {  // chunk tail: tiles 24..26
+      fp8x8 a[3], b[3];
+#pragma unroll
+      for (int u = 0; u < 3; ++u) {
+        a[u] = *reinterpret_cast<const fp8x8*>(ap + (kt0 + 24 + u) * 32);
+        b[u] = *reinterpret_cast<const fp8x8*>(bl + (24 + u) * 32);
+      }
+#pragma unroll
+      for (int u = 0; u < 3; ++u)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a[u], b[u], acc,
+                                                         0, 0, 0);
+    }
+    __syncthreads();
   }
   const int drow = mt * 16 + kgrp * 4;
   union {
