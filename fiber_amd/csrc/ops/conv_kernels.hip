@@ -317,6 +317,11 @@ conv_layer2(const __hip_bfloat16* __restrict__ wpert,
             unsigned char* __restrict__ act2) {
   __shared__ alignas(16) __hip_bfloat16 w2[C2][256];
   __shared__ float b2[C2];
+  // act1 staged in LDS: the 4x4-stride-2 im2col pattern touches each
+  // input pixel from up to 4 windows and BOTH M-tiles — ~7.7x read
+  // amplification that was going to L2/HBM (12.8 KB staged once
+  // instead; w2 16 KB + act1 12.8 KB still leaves 4 wg/CU).
+  __shared__ alignas(16) __hip_bfloat16 a1sh[O1 * O1 * C1];
   const int be = blockIdx.x;
   const int member = be / CENV;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
@@ -328,6 +333,9 @@ conv_layer2(const __hip_bfloat16* __restrict__ wpert,
     reinterpret_cast<bf16x8*>(&w2[0][0])[i] =
         reinterpret_cast<const bf16x8*>(wm + COFF_W2)[i];
   }
+  for (int i = tid; i < O1 * O1 * C1 / 8; i += blockDim.x)
+    reinterpret_cast<bf16x8*>(a1sh)[i] =
+        reinterpret_cast<const bf16x8*>(in)[i];
   if (tid < C2) b2[tid] = __bfloat162float(wm[COFF_B2 + tid]);
   __syncthreads();
 
@@ -353,7 +361,7 @@ conv_layer2(const __hip_bfloat16* __restrict__ wpert,
       const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
           &w2[arow & 31][kk * 32 + kgrp * 8]);
       const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-          &in[pix * C1 + (kgrp & 1) * 8]);
+          &a1sh[pix * C1 + (kgrp & 1) * 8]);
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0,
                                                     0, 0);
     }
