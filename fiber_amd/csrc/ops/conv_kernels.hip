@@ -6,9 +6,9 @@
 //       --fc 256--> --head 6-->
 //
 // Data layouts (all channel-last so im2col K-runs are memory-contiguous):
-//   obs  [member][env][y][x][ic=4]    bf16
+//   obs  [member][env][y][x][ic=4]    e4m3 (fp8)
 //   act1 [member][env][pos=20x20][16] bf16
-//   act2 [member][env][pos=9x9][32]   bf16   (flat k for fc = (y*9+x)*32+oc)
+//   act2 [member][env][pos=9x9][32]   e4m3 (flat k for fc = (y*9+x)*32+oc)
 //   act3 [member][env][256]           bf16
 // Weights: one flat fp32 master theta; es_perturb materializes per-member
 // bf16 perturbed copies (antithetic Philox pairs, same counter scheme as
@@ -240,6 +240,10 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
             __hip_bfloat16* __restrict__ act1) {
   __shared__ alignas(16) unsigned char w1[C1][256];
   __shared__ float b1[C1];
+  // obs staged in LDS: the 8x8-stride-4 im2col makes the 8 B fragment
+  // loads scattered (the kernel is latency-bound on them, not on
+  // bytes); one coalesced 28 KB stage turns them into LDS reads.
+  __shared__ alignas(16) unsigned char obsh[IMG * IMG * CIN];
   const int be = blockIdx.x;
   const int member = be / CENV;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
@@ -252,6 +256,9 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
         reinterpret_cast<const fp8x8*>(w1_fp8 +
                                        (size_t)member * (C1 * 256))[i];
   }
+  for (int i = tid; i < IMG * IMG * CIN / 16; i += blockDim.x)
+    reinterpret_cast<fam_uint4*>(obsh)[i] =
+        reinterpret_cast<const fam_uint4*>(ob)[i];
   if (tid < C1) b1[tid] = __bfloat162float(wm[COFF_B1 + tid]);
   __syncthreads();
 
@@ -280,9 +287,9 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {  // ky = kk
       const fp8x8 b0 = *reinterpret_cast<const fp8x8*>(
-          &ob[(((oy0 * 4 + kk) * IMG) + ox0 * 4 + kgrp * 2) * CIN]);
+          &obsh[(((oy0 * 4 + kk) * IMG) + ox0 * 4 + kgrp * 2) * CIN]);
       const fp8x8 b1f = *reinterpret_cast<const fp8x8*>(
-          &ob[(((oy1 * 4 + kk) * IMG) + ox1 * 4 + kgrp * 2) * CIN]);
+          &obsh[(((oy1 * 4 + kk) * IMG) + ox1 * 4 + kgrp * 2) * CIN]);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag[kk], b0,
                                                         acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag[kk], b1f,
