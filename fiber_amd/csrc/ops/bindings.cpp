@@ -36,12 +36,9 @@ extern "C" __global__ void conv_env_init(uint32_t, const uint32_t*, int,
                                          float*, float*);
 extern "C" __global__ void conv_noisegen(uint32_t, const uint32_t*,
                                          uint32_t, unsigned char*);
-extern "C" __global__ void conv_obsgen(const float*,
-                                       const __hip_bfloat16*,
-                                       const unsigned char*,
-                                       unsigned char*);
 extern "C" __global__ void conv_layer1(const __hip_bfloat16*,
-                                       const unsigned char*,
+                                       const unsigned char*, const float*,
+                                       const __hip_bfloat16*,
                                        const unsigned char*, int,
                                        __hip_bfloat16*);
 extern "C" __global__ void conv_layer2(const __hip_bfloat16*,
@@ -158,26 +155,18 @@ static void launch_conv_noisegen(uint32_t seed, uintptr_t iterp, uint32_t t,
   check(hipGetLastError(), "conv_noisegen launch");
 }
 
-static void launch_conv_obsgen(uintptr_t state, uintptr_t gtab,
-                               uintptr_t znoise, int nenv_total,
-                               uintptr_t obs, uintptr_t stream) {
-  hipLaunchKernelGGL(conv_obsgen, dim3(nenv_total), dim3(256), 0,
-                     (hipStream_t)stream, (const float*)state,
-                     (const __hip_bfloat16*)gtab,
-                     (const unsigned char*)znoise, (unsigned char*)obs);
-  check(hipGetLastError(), "conv_obsgen launch");
-}
-
 static void launch_conv_forward(uintptr_t wpert, uintptr_t w3_fp8,
-                                uintptr_t w1_fp8, uintptr_t obs,
+                                uintptr_t w1_fp8, uintptr_t state,
+                                uintptr_t gtab, uintptr_t znoise,
                                 uintptr_t act1, uintptr_t act2,
                                 uintptr_t act3, int nmembers,
                                 uintptr_t stream) {
   const int nenv = nmembers * 16;
   hipLaunchKernelGGL(conv_layer1, dim3(nenv), dim3(256), 0,
                      (hipStream_t)stream, (const __hip_bfloat16*)wpert,
-                     (const unsigned char*)w1_fp8,
-                     (const unsigned char*)obs, nenv,
+                     (const unsigned char*)w1_fp8, (const float*)state,
+                     (const __hip_bfloat16*)gtab,
+                     (const unsigned char*)znoise, nenv,
                      (__hip_bfloat16*)act1);
   check(hipGetLastError(), "conv_layer1 launch");
   hipLaunchKernelGGL(conv_layer2, dim3(nenv), dim3(256), 0,
@@ -240,13 +229,11 @@ PYBIND11_MODULE(_ops, m) {
   m.def("conv_noisegen", &launch_conv_noisegen, py::arg("seed"),
         py::arg("iterp"), py::arg("t"), py::arg("nenv"),
         py::arg("znoise"), py::arg("stream"));
-  m.def("conv_obsgen", &launch_conv_obsgen, py::arg("state"),
-        py::arg("gtab"), py::arg("znoise"), py::arg("nenv_total"),
-        py::arg("obs"), py::arg("stream"));
   m.def("conv_forward", &launch_conv_forward, py::arg("wpert"),
-        py::arg("w3_fp8"), py::arg("w1_fp8"), py::arg("obs"),
-        py::arg("act1"), py::arg("act2"), py::arg("act3"),
-        py::arg("nmembers"), py::arg("stream"));
+        py::arg("w3_fp8"), py::arg("w1_fp8"), py::arg("state"),
+        py::arg("gtab"), py::arg("znoise"), py::arg("act1"),
+        py::arg("act2"), py::arg("act3"), py::arg("nmembers"),
+        py::arg("stream"));
   m.def("conv_head_env", &launch_conv_head_env, py::arg("wpert"),
         py::arg("act3"), py::arg("nmembers"), py::arg("env_A"),
         py::arg("env_B"), py::arg("state"), py::arg("racc"),

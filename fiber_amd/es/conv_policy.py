@@ -114,10 +114,6 @@ class ConvESEngine:
                                   device=device)
         self.w1_fp8 = torch.empty(pop, 16 * 256, dtype=torch.uint8,
                                   device=device)
-        # pixel observations also ride fp8 (obsgen/conv1 were ~45% of the
-        # step and BW/latency-bound on the 84x84x4 obs buffers)
-        self.obs = torch.empty(pop * E, 84 * 84 * 4, dtype=torch.uint8,
-                               device=device)
         self.act1 = torch.empty(pop * E, 20 * 20 * 16, dtype=bf,
                                 device=device)
         self.act2 = torch.empty(pop * E, 2592, dtype=torch.uint8,
@@ -146,7 +142,7 @@ class ConvESEngine:
         # shared obs-noise staging, one e4m3 field per stream chunk: the
         # noise is keyed by (env, pos, t) only (common random numbers
         # across the population), so conv_noisegen fills this once per
-        # (chunk, t) and conv_obsgen re-reads it for every member —
+        # (chunk, t) and conv_layer1 re-reads it for every member —
         # 1/(pop/chunk) of the philox work the fused kernel redid.
         # Chunks run at skewed timesteps, hence one buffer each.
         self.znoise = torch.empty(nhalves, E * 84 * 84 * 4,
@@ -172,7 +168,6 @@ class ConvESEngine:
         wpert = self.wpert[m0:].data_ptr()
         w3_fp8 = self.w3_fp8[m0:].data_ptr()
         w1_fp8 = self.w1_fp8[m0:].data_ptr()
-        obs = self.obs[m0 * E:].data_ptr()
         act1 = self.act1[m0 * E:].data_ptr()
         act2 = self.act2[m0 * E:].data_ptr()
         act3 = self.act3[m0 * E:].data_ptr()
@@ -185,10 +180,11 @@ class ConvESEngine:
         znoise = self.znoise[half].data_ptr()
         for t in range(cfg.horizon):
             o.conv_noisegen(cfg.seed, iterp, t, E, znoise, stream)
-            o.conv_obsgen(state, self.gtab_bf.data_ptr(), znoise,
-                          pop * E, obs, stream)
-            o.conv_forward(wpert, w3_fp8, w1_fp8, obs, act1, act2, act3,
-                           pop, stream)
+            # obs are generated INSIDE conv_layer1's LDS staging pass —
+            # the 84x84x4 image never exists in HBM
+            o.conv_forward(wpert, w3_fp8, w1_fp8, state,
+                           self.gtab_bf.data_ptr(), znoise, act1, act2,
+                           act3, pop, stream)
             o.conv_head_env(wpert, act3, pop, self.env_A.data_ptr(),
                             self.env_B.data_ptr(), state, racc, stream)
 
