@@ -76,6 +76,17 @@ def gpu_count():
 
 def Pool(processes=None, initializer=None, initargs=(), maxtasksperchild=None,
          error_handling=False, **kwargs):
+    """Pool factory (reference ``fiber/context.py:38-45`` semantics).
+
+    ``error_handling=False`` (default, same as the reference context
+    factory) returns :class:`~fiber_amd.pool.ZPool`: worker exceptions
+    are delivered to the caller.  ``error_handling=True`` returns
+    :class:`~fiber_amd.pool.ResilientZPool` (also exposed as
+    ``fiber_amd.pool.Pool``, matching the reference's module alias):
+    a failing worker is killed and its pending chunks are resubmitted —
+    tasks must be idempotent.  The two entry points deliberately differ,
+    exactly as in the reference.
+    """
     from . import pool as _pool_mod
 
     if error_handling:

@@ -25,7 +25,11 @@ def _visible_devices():
         "CUDA_VISIBLE_DEVICES"
     )
     if env:
-        return list(range(len(env.split(","))))
+        # The child's HIP_VISIBLE_DEVICES *replaces* (does not nest
+        # within) this process's, so hand out the parent's actual
+        # ordinals — returning range(len(...)) would pin children to
+        # physical GPUs outside the parent's allocation.
+        return [int(x) for x in env.split(",") if x.strip() != ""]
     try:
         import torch
 

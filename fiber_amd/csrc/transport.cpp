@@ -43,6 +43,7 @@
 #include <string>
 #include <vector>
 
+#include <dirent.h>
 #include <fcntl.h>
 #include <linux/futex.h>
 #include <pthread.h>
@@ -57,12 +58,13 @@ namespace py = pybind11;
 
 namespace {
 
-constexpr uint32_t kMagic = 0xFA3B71A7u;  // bumped: futex-word header
+constexpr uint32_t kMagic = 0xFA3B71A8u;  // bumped: spill-flag header
 constexpr uint32_t kWrapMarker = 0xFFFFFFFFu;
 constexpr size_t kAlign = 8;
 
 constexpr uint32_t kStReserved = 0u;
 constexpr uint32_t kStCommitted = 1u;
+constexpr uint32_t kFlagSpill = 1u;  // record flags word (header byte 12)
 constexpr size_t kRecHdr = 16;
 constexpr size_t kMaxBatch = 128;  // records per lock hold in batched ops
 
@@ -153,6 +155,99 @@ struct WaiterScope {
   ~WaiterScope() { as_atomic(w)->fetch_sub(1, std::memory_order_seq_cst); }
 };
 
+// ---------------------------------------------------------------------------
+// Spill segments: a message too big for the ring rides a one-shot shm
+// segment; the ring carries a small control record (kFlagSpill) naming it.
+// The reader copies the payload out and unlinks the segment; the ring
+// owner sweeps stale "<ring>.sp.*" segments on create and unlink (leaks
+// only happen when a writer dies between creating the segment and
+// committing the control record).
+// ---------------------------------------------------------------------------
+
+// Control payload: [u64 real_len][u16 name_len][name bytes]
+inline size_t spill_ctrl_encode(char* out, uint64_t real_len,
+                                const char* name, size_t name_len) {
+  std::memcpy(out, &real_len, 8);
+  uint16_t nl = (uint16_t)name_len;
+  std::memcpy(out + 8, &nl, 2);
+  std::memcpy(out + 10, name, name_len);
+  return 10 + name_len;
+}
+
+inline bool spill_ctrl_decode(const char* ctrl, size_t ctrl_len,
+                              uint64_t* real_len, std::string* name) {
+  if (ctrl_len < 10) return false;
+  std::memcpy(real_len, ctrl, 8);
+  uint16_t nl;
+  std::memcpy(&nl, ctrl + 8, 2);
+  if (ctrl_len < (size_t)10 + nl) return false;
+  name->assign(ctrl + 10, nl);
+  return true;
+}
+
+// Writes `len` bytes into a fresh exclusive shm segment; returns its name.
+inline std::string spill_write(const std::string& ring_name, const char* buf,
+                               size_t len) {
+  static std::atomic<uint64_t> counter{0};
+  char nm[192];
+  snprintf(nm, sizeof(nm), "%s.sp.%u.%llu", ring_name.c_str(),
+           (unsigned)getpid(),
+           (unsigned long long)counter.fetch_add(1));
+  int fd = shm_open(nm, O_CREAT | O_EXCL | O_RDWR, 0600);
+  if (fd < 0)
+    throw std::runtime_error(std::string("spill shm_open failed: ") + nm);
+  size_t total = len ? len : 1;
+  if (ftruncate(fd, (off_t)total) != 0) {
+    close(fd);
+    shm_unlink(nm);
+    throw std::runtime_error(std::string("spill ftruncate failed: ") + nm);
+  }
+  void* mem = mmap(nullptr, total, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
+  close(fd);
+  if (mem == MAP_FAILED) {
+    shm_unlink(nm);
+    throw std::runtime_error(std::string("spill mmap failed: ") + nm);
+  }
+  if (len) std::memcpy(mem, buf, len);
+  munmap(mem, total);
+  return std::string(nm);
+}
+
+// Copies the spill payload into dst and unlinks the segment.
+inline void spill_read(const std::string& nm, char* dst, uint64_t len) {
+  int fd = shm_open(nm.c_str(), O_RDONLY, 0600);
+  if (fd < 0)
+    throw std::runtime_error("spill segment missing: " + nm);
+  if (len) {
+    void* mem = mmap(nullptr, len, PROT_READ, MAP_SHARED, fd, 0);
+    close(fd);
+    if (mem == MAP_FAILED) {
+      shm_unlink(nm.c_str());
+      throw std::runtime_error("spill mmap(read) failed: " + nm);
+    }
+    std::memcpy(dst, mem, len);
+    munmap(mem, len);
+  } else {
+    close(fd);
+  }
+  shm_unlink(nm.c_str());
+}
+
+// Unlink every "<ring>.sp.*" segment (stale leftovers of crashed writers).
+inline void spill_sweep(const std::string& ring_name) {
+  DIR* d = opendir("/dev/shm");
+  if (!d) return;
+  std::string prefix = ring_name + ".sp.";
+  struct dirent* e;
+  std::vector<std::string> victims;
+  while ((e = readdir(d)) != nullptr) {
+    if (strncmp(e->d_name, prefix.c_str(), prefix.size()) == 0)
+      victims.emplace_back(e->d_name);
+  }
+  closedir(d);
+  for (auto& v : victims) shm_unlink(v.c_str());
+}
+
 class RobustLock {
  public:
   explicit RobustLock(pthread_mutex_t* mu) : mu_(mu) {
@@ -183,10 +278,16 @@ class ShmRing {
   ShmRing(const std::string& name, bool create, size_t capacity,
           double open_timeout)
       : name_(name), owner_(create) {
+    // The record format assumes every offset is kAlign-aligned; an
+    // unaligned user-configured capacity could leave < 4 bytes for the
+    // wrap marker at the end of the data area.  Round down + floor.
+    capacity &= ~(kAlign - 1);
+    if (capacity < 4096) capacity = 4096;
     size_t total = sizeof(Header) + capacity;
     int fd = -1;
     if (create) {
       shm_unlink(name.c_str());  // stale segment from a crashed run
+      spill_sweep(name);         // + stale spill segments
       fd = shm_open(name.c_str(), O_CREAT | O_EXCL | O_RDWR, 0600);
       if (fd < 0)
         throw std::runtime_error("shm_open create failed: " + name);
@@ -262,9 +363,30 @@ class ShmRing {
     }
   }
 
+  // Spill policy: a message that cannot fit, or would occupy more than a
+  // quarter of the ring (starving other producers), rides a spill segment.
+  bool needs_spill(size_t len) const {
+    size_t need = record_bytes(len);
+    uint64_t cap = hdr_->capacity;
+    return need + kRecHdr + kAlign >= cap || need > cap / 4;
+  }
+
   // timeout < 0: block forever; timeout == 0: non-blocking.
   // Returns false on timeout; throws if the ring is closed.
   bool send(const char* buf, size_t len, double timeout) {
+    if (needs_spill(len)) {
+      std::string nm = spill_write(name_, buf, len);
+      char ctrl[224];
+      size_t clen = spill_ctrl_encode(ctrl, len, nm.c_str(), nm.size());
+      bool ok = send_record(ctrl, clen, kFlagSpill, timeout);
+      if (!ok) shm_unlink(nm.c_str());  // timed out: reclaim the segment
+      return ok;
+    }
+    return send_record(buf, len, 0, timeout);
+  }
+
+  bool send_record(const char* buf, size_t len, uint32_t flags,
+                   double timeout) {
     size_t need = record_bytes(len);
     if (need + kRecHdr + kAlign >= hdr_->capacity)
       throw std::runtime_error("message larger than ring capacity");
@@ -307,6 +429,7 @@ class ShmRing {
           as_atomic(reinterpret_cast<uint32_t*>(data_ + rec + 4))
               ->store(kStReserved, std::memory_order_relaxed);
           std::memcpy(data_ + rec + 8, &pid, 4);
+          std::memcpy(data_ + rec + 12, &flags, 4);
           hdr_->tail = (tail + need) % cap;
           hdr_->used += need;
           hdr_->msg_count += 1;
@@ -344,6 +467,36 @@ class ShmRing {
   // enqueued (< n only on timeout / non-blocking backpressure).
   size_t send_many(const char* const* bufs, const size_t* lens, size_t n,
                    double timeout) {
+    // Pre-spill oversized messages OUTSIDE the lock: each becomes a small
+    // flagged control record; segments of messages that never get sent
+    // (timeout) are reclaimed below.
+    std::vector<const char*> xbufs(bufs, bufs + n);
+    std::vector<size_t> xlens(lens, lens + n);
+    std::vector<uint32_t> flags(n, 0);
+    std::vector<std::string> ctrl_store(n);
+    std::vector<std::string> seg_names(n);
+    for (size_t i = 0; i < n; ++i) {
+      if (needs_spill(lens[i])) {
+        seg_names[i] = spill_write(name_, bufs[i], lens[i]);
+        ctrl_store[i].resize(224);
+        size_t clen = spill_ctrl_encode(&ctrl_store[i][0], lens[i],
+                                        seg_names[i].c_str(),
+                                        seg_names[i].size());
+        ctrl_store[i].resize(clen);
+        xbufs[i] = ctrl_store[i].data();
+        xlens[i] = clen;
+        flags[i] = kFlagSpill;
+      }
+    }
+    size_t done = send_many_records(xbufs.data(), xlens.data(),
+                                    flags.data(), n, timeout);
+    for (size_t i = done; i < n; ++i)
+      if (flags[i]) shm_unlink(seg_names[i].c_str());
+    return done;
+  }
+
+  size_t send_many_records(const char* const* bufs, const size_t* lens,
+                           const uint32_t* flags, size_t n, double timeout) {
     const double deadline = timeout > 0 ? monotonic_now() + timeout : 0.0;
     size_t done = 0;
     uint64_t recs[kMaxBatch];
@@ -379,6 +532,7 @@ class ShmRing {
             as_atomic(reinterpret_cast<uint32_t*>(data_ + tail + 4))
                 ->store(kStReserved, std::memory_order_relaxed);
             std::memcpy(data_ + tail + 8, &pid, 4);
+            std::memcpy(data_ + tail + 12, &flags[done + burst], 4);
             recs[burst] = tail;
             hdr_->tail = (tail + need) % cap;
             hdr_->used += need;
@@ -416,7 +570,8 @@ class ShmRing {
   // Consumer-side helper: with the lock held, resolve the head record.
   // Returns: 0 = a committed record is ready (out params set);
   //          1 = ring empty; 2 = head reserved (writer mid-copy).
-  int resolve_head(uint64_t* head_out, uint32_t* len_out) {
+  int resolve_head(uint64_t* head_out, uint32_t* len_out,
+                   uint32_t* flags_out) {
     if (hdr_->msg_count == 0) return 1;
     uint64_t cap = hdr_->capacity;
     uint64_t head = hdr_->head;
@@ -432,7 +587,27 @@ class ShmRing {
             ->load(std::memory_order_acquire);
     *head_out = head;
     *len_out = len32;
+    std::memcpy(flags_out, data_ + head + 12, 4);
     return state == kStCommitted ? 0 : 2;
+  }
+
+  // With the lock held: copy a committed spill record's payload into
+  // *out and consume it (the control record names the segment).
+  void consume_spill(uint64_t head, uint32_t len32, std::string* out) {
+    uint64_t real_len;
+    std::string nm;
+    if (!spill_ctrl_decode(data_ + head + kRecHdr, len32, &real_len, &nm)) {
+      consume_record(head, len32);
+      throw std::runtime_error("corrupt spill control record");
+    }
+    out->resize(real_len);
+    try {
+      spill_read(nm, real_len ? &(*out)[0] : nullptr, real_len);
+    } catch (...) {
+      consume_record(head, len32);
+      throw;
+    }
+    consume_record(head, len32);
   }
 
   void consume_record(uint64_t head, uint32_t len32) {
@@ -451,7 +626,7 @@ class ShmRing {
   }
 
   // Shared blocking structure for recv / recv_into / peek.
-  // op(head, len) -> true when it consumed / is satisfied.
+  // op(head, len, flags) -> true when it consumed / is satisfied.
   template <typename Op>
   bool recv_loop(double timeout, Op&& op) {
     const double deadline =
@@ -465,9 +640,10 @@ class ShmRing {
         RobustLock lock(&hdr_->mu);
         uint64_t head;
         uint32_t len32;
-        st = resolve_head(&head, &len32);
+        uint32_t flags;
+        st = resolve_head(&head, &len32, &flags);
         if (st == 0) {
-          if (op(head, len32)) {
+          if (op(head, len32, flags)) {
             lock.unlock();
             bump_and_wake(&hdr_->fut_not_full, &hdr_->waiters_not_full);
             return true;
@@ -536,13 +712,19 @@ class ShmRing {
         RobustLock lock(&hdr_->mu);
         uint64_t head;
         uint32_t len32;
-        st = resolve_head(&head, &len32);
+        uint32_t flags;
+        st = resolve_head(&head, &len32, &flags);
         if (st == 0) {
           do {
-            out->emplace_back(data_ + head + kRecHdr, len32);
-            consume_record(head, len32);
+            if (flags & kFlagSpill) {
+              out->emplace_back();
+              consume_spill(head, len32, &out->back());
+            } else {
+              out->emplace_back(data_ + head + kRecHdr, len32);
+              consume_record(head, len32);
+            }
           } while (out->size() < max_n &&
-                   resolve_head(&head, &len32) == 0);
+                   resolve_head(&head, &len32, &flags) == 0);
           lock.unlock();
           bump_and_wake(&hdr_->fut_not_full, &hdr_->waiters_not_full);
           return out->size();
@@ -586,27 +768,55 @@ class ShmRing {
   }
 
   bool recv(std::string* out, double timeout) {
-    return recv_loop(timeout, [&](uint64_t head, uint32_t len32) {
-      out->assign(data_ + head + kRecHdr, len32);
-      consume_record(head, len32);
-      return true;
-    });
+    return recv_loop(
+        timeout, [&](uint64_t head, uint32_t len32, uint32_t flags) {
+          if (flags & kFlagSpill) {
+            consume_spill(head, len32, out);
+            return true;
+          }
+          out->assign(data_ + head + kRecHdr, len32);
+          consume_record(head, len32);
+          return true;
+        });
   }
 
   // Single-copy receive.  Returns len, -1 on timeout, or -(len)-2 if the
   // buffer is too small (message left in place).
   int64_t recv_into(char* buf, size_t buflen, double timeout) {
     int64_t result = -1;
-    bool ok = recv_loop(timeout, [&](uint64_t head, uint32_t len32) {
-      if ((size_t)len32 > buflen) {
-        result = -((int64_t)len32) - 2;
-        return false;
-      }
-      std::memcpy(buf, data_ + head + kRecHdr, len32);
-      consume_record(head, len32);
-      result = (int64_t)len32;
-      return true;
-    });
+    bool ok = recv_loop(
+        timeout, [&](uint64_t head, uint32_t len32, uint32_t flags) {
+          if (flags & kFlagSpill) {
+            uint64_t real_len;
+            std::string nm;
+            if (!spill_ctrl_decode(data_ + head + kRecHdr, len32,
+                                   &real_len, &nm)) {
+              consume_record(head, len32);
+              throw std::runtime_error("corrupt spill control record");
+            }
+            if (real_len > buflen) {
+              result = -((int64_t)real_len) - 2;
+              return false;  // left in place (incl. the segment)
+            }
+            try {
+              spill_read(nm, buf, real_len);
+            } catch (...) {
+              consume_record(head, len32);
+              throw;
+            }
+            consume_record(head, len32);
+            result = (int64_t)real_len;
+            return true;
+          }
+          if ((size_t)len32 > buflen) {
+            result = -((int64_t)len32) - 2;
+            return false;
+          }
+          std::memcpy(buf, data_ + head + kRecHdr, len32);
+          consume_record(head, len32);
+          result = (int64_t)len32;
+          return true;
+        });
     (void)ok;
     return result;
   }
@@ -614,10 +824,21 @@ class ShmRing {
   // Size of the next committed message, or -1 on timeout.  No consume.
   int64_t peek_size(double timeout) {
     int64_t result = -1;
-    recv_loop(timeout, [&](uint64_t /*head*/, uint32_t len32) {
-      result = (int64_t)len32;
-      return true;
-    });
+    recv_loop(timeout,
+              [&](uint64_t head, uint32_t len32, uint32_t flags) {
+                if (flags & kFlagSpill) {
+                  uint64_t real_len;
+                  std::string nm;
+                  if (spill_ctrl_decode(data_ + head + kRecHdr, len32,
+                                        &real_len, &nm))
+                    result = (int64_t)real_len;
+                  else
+                    result = (int64_t)len32;
+                  return true;
+                }
+                result = (int64_t)len32;
+                return true;
+              });
     return result;
   }
 
@@ -643,7 +864,10 @@ class ShmRing {
     futex_wake_all(&hdr_->fut_not_full);
   }
 
-  void unlink_ring() { shm_unlink(name_.c_str()); }
+  void unlink_ring() {
+    shm_unlink(name_.c_str());
+    spill_sweep(name_);  // unread / leaked spill segments die with the ring
+  }
 
   uint64_t size() const { return hdr_ ? hdr_->msg_count : 0; }
   uint64_t total_in() const { return hdr_ ? hdr_->total_in : 0; }

@@ -320,11 +320,28 @@ class AsyncProxyResult:
         self._value = None
         self._error = None
 
-    def _fill_next(self):
+    def _fill_next(self, timeout=None):
+        import select
+        import time as _time
+
         conn = self._proxy._conn
         pending = self._proxy._pending
+        deadline = None if timeout is None else _time.monotonic() + timeout
         with conn.lock:
             while pending and not self._done:
+                if deadline is not None:
+                    remaining = deadline - _time.monotonic()
+                    if remaining <= 0:
+                        raise TimeoutError("async result timed out")
+                    # Wait for data outside recv_msg so a hung server
+                    # cannot block past the deadline (a timeout mid-frame
+                    # would corrupt the stream; select-then-recv keeps the
+                    # framing intact).
+                    ready, _, _ = select.select(
+                        [conn.ensure()], [], [], remaining
+                    )
+                    if not ready:
+                        raise TimeoutError("async result timed out")
                 head = pending[0]
                 try:
                     value = conn.recv_only()
@@ -336,7 +353,7 @@ class AsyncProxyResult:
 
     def get(self, timeout=None):
         if not self._done:
-            self._fill_next()
+            self._fill_next(timeout)
         if self._error is not None:
             raise self._error
         return self._value

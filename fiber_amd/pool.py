@@ -452,7 +452,9 @@ def _pool_worker_core(
         seq, base, func_blob, args, star, kwds = task
         if seq == _SENTINEL_SEQ:
             break
-        key = hash(func_blob)
+        # Keyed by the blob bytes themselves: a hash collision between two
+        # distinct pickled functions must not silently run the wrong one.
+        key = bytes(func_blob)
         func = func_cache.get(key)
         if func is None:
             func = serialization.loads(func_blob)
@@ -842,8 +844,17 @@ class ZPool:
                 proc.terminate()
         self._inventory.fail_all(RuntimeError("pool terminated"))
 
-    def join(self, timeout=30.0):
-        deadline = time.monotonic() + timeout
+    def join(self, timeout=None):
+        """Wait for all workers to exit.
+
+        stdlib semantics: after a plain ``close()`` this blocks until the
+        workers drain their remaining tasks and exit, however long that
+        takes (``timeout=None``).  Pass a number to cap the wait (a pool
+        abandoned at the cap is still torn down safely)."""
+        if self._state == "run":
+            raise ValueError("Pool is still running; call close() or "
+                             "terminate() before join()")
+        deadline = None if timeout is None else time.monotonic() + timeout
         # Freeze the population first: once the maintainer thread has
         # exited, no new worker can be spawned behind our back.
         if self._worker_thread is not None:
@@ -856,7 +867,10 @@ class ZPool:
         for proc in workers:
             if proc._popen is None:
                 continue  # registered but never started (shutdown race)
-            proc.join(max(0.1, deadline - time.monotonic()))
+            if deadline is None:
+                proc.join()
+            else:
+                proc.join(max(0.1, deadline - time.monotonic()))
         with self._worker_lock:
             self._workers.clear()
         if self._state != "terminated":

@@ -64,10 +64,15 @@ class Connection:
         return data
 
     def poll(self, timeout=0.0):
-        """True if a message is available within *timeout* seconds."""
+        """True if a message is available within *timeout* seconds.
+
+        ``poll(None)`` blocks until a message arrives (stdlib
+        ``multiprocessing.Connection.poll`` semantics); ``poll()`` /
+        ``poll(0)`` is non-blocking."""
         if self._buffered is not None:
             return True
-        data = self._ensure().recv(timeout if timeout else 0.0)
+        ring_timeout = -1.0 if timeout is None else float(timeout)
+        data = self._ensure().recv(ring_timeout)
         if data is None:
             return False
         self._buffered = data

@@ -25,6 +25,16 @@ def _random_error(x):
     return x * 2
 
 
+def _checksum_bytes(a):
+    return int(a.astype("u8").sum())
+
+
+def _make_big_array(mb):
+    import numpy as np
+
+    return np.full(mb << 20, 7, dtype=np.uint8)
+
+
 def _slow_identity(x):
     time.sleep(0.01)
     return x
@@ -111,6 +121,19 @@ class TestZPool:
     def test_context_manager(self):
         with ZPool(processes=2) as p:
             assert p.map(_square, range(10)) == [x * x for x in range(10)]
+
+    def test_large_payloads_roundtrip(self):
+        """Task args/results far past the ring capacity (8 MB default)
+        ride the spill path: a 64 MB array round-trips unchanged."""
+        import numpy as np
+
+        with ZPool(processes=2) as p:
+            arrs = [np.full(16 << 20, i, dtype=np.uint8) for i in range(3)]
+            out = p.map(_checksum_bytes, arrs, chunksize=1)
+            assert out == [int(a[0]) * len(a) for a in arrs]
+            big = p.apply(_make_big_array, (64,))
+            assert big.nbytes == 64 << 20
+            assert big[0] == 7 and big[-1] == 7
 
 
 class TestResilientZPool:

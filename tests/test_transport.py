@@ -1,10 +1,16 @@
 """Shm-ring transport engine tests."""
 
+import glob
+import os
 import threading
 
 import pytest
 
 from fiber_amd.transport import ShmRing, Socket, new_address
+
+
+def _spill_segments(ring_name):
+    return glob.glob("/dev/shm/%s.sp.*" % ring_name)
 
 
 class TestShmRing:
@@ -45,12 +51,78 @@ class TestShmRing:
             ring.close()
             ring.unlink()
 
-    def test_oversize_message_raises(self):
+    def test_oversize_message_spills_and_roundtrips(self):
+        """A message past ring capacity rides a spill segment: it still
+        round-trips byte-exact and leaves no segment behind."""
         name = new_address("fam-t")
         ring = ShmRing(name, True, 4 << 10, 5.0)
         try:
-            with pytest.raises(RuntimeError):
-                ring.send(b"y" * (8 << 10), 1.0)
+            for size in (8 << 10, 1 << 20, 10 << 20):
+                msg = os.urandom(size)
+                assert ring.send(msg, 5.0)
+                assert ring.recv(5.0) == msg
+            assert not _spill_segments(name)
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_spill_mixed_with_inline_in_order(self):
+        """FIFO order is preserved across inline records and spill
+        control records, including through the batched ops."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 8 << 10, 5.0)
+        try:
+            msgs = [b"s" * 100, os.urandom(64 << 10), b"t" * 200,
+                    os.urandom(3 << 20), b"u" * 50]
+            assert ring.send_many(msgs, 5.0) == len(msgs)
+            got = list(ring.recv_many(3, 5.0))
+            got += [ring.recv(5.0) for _ in range(len(msgs) - len(got))]
+            assert got == msgs
+            assert not _spill_segments(name)
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_spill_peek_and_recv_into(self):
+        """peek_size reports the REAL payload size of a spilled message;
+        recv_into with a small buffer declines without consuming."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        try:
+            msg = os.urandom(256 << 10)
+            assert ring.send(msg, 5.0)
+            assert ring.peek_size(1.0) == len(msg)
+            small = bytearray(1024)
+            n = ring.recv_into(small, 1.0)
+            assert n == -(len(msg)) - 2  # too small, left in place
+            big = bytearray(len(msg))
+            assert ring.recv_into(big, 1.0) == len(msg)
+            assert bytes(big) == msg
+            assert not _spill_segments(name)
+        finally:
+            ring.close()
+            ring.unlink()
+
+    def test_spill_sweep_on_unlink(self):
+        """Unread spill segments are swept when the ring owner unlinks."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 4 << 10, 5.0)
+        ring.send(os.urandom(64 << 10), 5.0)
+        ring.send(os.urandom(64 << 10), 5.0)
+        assert len(_spill_segments(name)) == 2
+        ring.close()
+        ring.unlink()
+        assert not _spill_segments(name)
+
+    def test_unaligned_capacity_is_floored(self):
+        """A capacity that is not a multiple of 8 is rounded down (the
+        wrap marker needs 4 bytes before the end of the data area)."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, (4 << 10) + 5, 5.0)
+        try:
+            for _ in range(64):  # drift across the wrap point
+                ring.send(b"z" * 500, 1.0)
+                assert ring.recv(1.0) == b"z" * 500
         finally:
             ring.close()
             ring.unlink()

@@ -11,7 +11,10 @@ from fiber_amd.transport import ShmRing, new_address
 
 CAP = 4096  # small: forces wraps + backpressure in nearly every run
 
-payloads = st.binary(min_size=0, max_size=1200)
+# max_size deliberately exceeds CAP: payloads past capacity/4 ride the
+# spill path (one-shot shm segments), so the model covers inline records,
+# wrap markers AND spill control records in the same FIFO.
+payloads = st.binary(min_size=0, max_size=9000)
 ops = st.lists(
     st.one_of(
         st.tuples(st.just("send"), payloads),
