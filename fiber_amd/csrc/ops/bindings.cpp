@@ -8,6 +8,7 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
+#include <rocprim/rocprim.hpp>
 
 #include <stdexcept>
 #include <string>
@@ -23,6 +24,9 @@ extern "C" __global__ void es_rollout_mlp(
 extern "C" __global__ void es_grad(const float*, int, int, int, uint32_t,
                                    uint32_t, int, float*);
 extern "C" __global__ void centered_rank(const float*, int, float*);
+extern "C" __global__ void rank_pack_keys(const float*, int, unsigned*,
+                                          unsigned*);
+extern "C" __global__ void rank_scatter(const unsigned*, int, float*);
 extern "C" __global__ void mlp_policy_forward(const float*, const float*,
                                               int, float*);
 extern "C" __global__ void mfma_gemm64_probe(const float*, const float*,
@@ -99,6 +103,51 @@ static void launch_centered_rank(uintptr_t f, int n, uintptr_t out,
   hipLaunchKernelGGL(centered_rank, dim3(bx), dim3(256), 0,
                      (hipStream_t)stream, (const float*)f, n, (float*)out);
   check(hipGetLastError(), "centered_rank launch");
+}
+
+// Sort-based rank for large populations: pack order-preserving uint keys,
+// stable device radix sort (rocPRIM — AMD-native), scatter ranks.
+// Workspace layout: [keys_in][keys_out][vals_in][vals_out][rocprim temp],
+// each array 256-byte aligned; caller allocates (torch caching allocator).
+static inline uint64_t rank_array_stride(int n) {
+  return (((uint64_t)n * 4) + 255) & ~(uint64_t)255;
+}
+
+static uint64_t centered_rank_sorted_workspace(int n) {
+  size_t temp_bytes = 0;
+  check(rocprim::radix_sort_pairs(nullptr, temp_bytes,
+                                  (const unsigned*)nullptr,
+                                  (unsigned*)nullptr,
+                                  (const unsigned*)nullptr,
+                                  (unsigned*)nullptr, (size_t)n),
+        "radix_sort_pairs size query");
+  return 4 * rank_array_stride(n) + (uint64_t)temp_bytes + 256;
+}
+
+static void launch_centered_rank_sorted(uintptr_t f, int n, uintptr_t out,
+                                        uintptr_t work, uint64_t work_bytes,
+                                        uintptr_t stream) {
+  const uint64_t stride = rank_array_stride(n);
+  if (work_bytes < 4 * stride)
+    throw std::runtime_error("centered_rank_sorted workspace too small");
+  unsigned* keys_in = (unsigned*)work;
+  unsigned* keys_out = (unsigned*)(work + stride);
+  unsigned* vals_in = (unsigned*)(work + 2 * stride);
+  unsigned* vals_out = (unsigned*)(work + 3 * stride);
+  void* temp = (void*)(work + 4 * stride);
+  size_t temp_bytes = (size_t)(work_bytes - 4 * stride);
+  const int bx = (n + 255) / 256;
+  hipLaunchKernelGGL(rank_pack_keys, dim3(bx), dim3(256), 0,
+                     (hipStream_t)stream, (const float*)f, n, keys_in,
+                     vals_in);
+  check(hipGetLastError(), "rank_pack_keys launch");
+  check(rocprim::radix_sort_pairs(temp, temp_bytes, keys_in, keys_out,
+                                  vals_in, vals_out, (size_t)n, 0, 32,
+                                  (hipStream_t)stream),
+        "radix_sort_pairs");
+  hipLaunchKernelGGL(rank_scatter, dim3(bx), dim3(256), 0,
+                     (hipStream_t)stream, vals_out, n, (float*)out);
+  check(hipGetLastError(), "rank_scatter launch");
 }
 
 static void launch_mlp_forward(uintptr_t theta, uintptr_t x, int batch,
@@ -209,6 +258,11 @@ PYBIND11_MODULE(_ops, m) {
         py::arg("nparams"), py::arg("grad"), py::arg("stream"));
   m.def("centered_rank", &launch_centered_rank, py::arg("f"), py::arg("n"),
         py::arg("out"), py::arg("stream"));
+  m.def("centered_rank_sorted_workspace", &centered_rank_sorted_workspace,
+        py::arg("n"));
+  m.def("centered_rank_sorted", &launch_centered_rank_sorted, py::arg("f"),
+        py::arg("n"), py::arg("out"), py::arg("work"),
+        py::arg("work_bytes"), py::arg("stream"));
   m.def("mlp_policy_forward", &launch_mlp_forward, py::arg("theta"),
         py::arg("x"), py::arg("batch"), py::arg("logits"),
         py::arg("stream"));

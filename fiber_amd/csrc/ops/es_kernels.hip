@@ -406,6 +406,31 @@ extern "C" __global__ void centered_rank(const float* __restrict__ f, int n,
 }
 
 // ---------------------------------------------------------------------------
+// Sort-based centered rank for large populations (>16k members, e.g. the
+// named 8-GPU config at pop 131,072, where the O(n^2) kernel would cost
+// milliseconds).  Keys are floats mapped to order-preserving uint32
+// (sign-flip trick); a stable device radix sort (rocPRIM, driven from the
+// host launcher) then yields rank = sorted position with the same
+// index-order tie-break as torch.argsort(stable=True).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void rank_pack_keys(const float* __restrict__ f, int n,
+                                          unsigned* __restrict__ keys,
+                                          unsigned* __restrict__ vals) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const unsigned u = __float_as_uint(f[i]);
+  keys[i] = (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+  vals[i] = (unsigned)i;
+}
+
+extern "C" __global__ void rank_scatter(const unsigned* __restrict__ vals,
+                                        int n, float* __restrict__ out) {
+  const int k = blockIdx.x * blockDim.x + threadIdx.x;
+  if (k >= n) return;
+  out[vals[k]] = (float)k / (float)(n - 1) - 0.5f;
+}
+
+// ---------------------------------------------------------------------------
 // Standalone batched policy forward (numerics target + serving op):
 // logits[b][ACT] for X[b][OBS], unperturbed theta.  One workgroup per
 // 64-row batch tile; same MFMA path as the rollout kernel.

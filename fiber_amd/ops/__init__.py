@@ -84,19 +84,31 @@ def es_grad(wpair, pair_begin, pair_end, seed, iteration, device,
     return grad
 
 
+_rank_workspace = {}  # device -> (n, uint8 tensor); grow-only cache
+
+
 def centered_rank(fitness):
     """Centered rank transform in [-0.5, 0.5].
 
     The O(n^2) comparison kernel wins for small populations (one launch,
-    no sort); past ~16k members the argsort path (device radix sort) is
-    asymptotically required — at 64k members the n^2 kernel would cost
-    milliseconds."""
+    no sort); past 16k members (e.g. the named 8-GPU config at pop
+    131,072) a stable device radix sort (rocPRIM) + scatter takes over —
+    identical output incl. the index-order tie-break."""
     _check(fitness, "fitness")
     n = fitness.numel()
-    if n > 16384:
-        return centered_rank_ref(fitness)
     ops = _require_ops()
     out = torch.empty_like(fitness)
+    if n > 16384:
+        dev = fitness.device
+        cached = _rank_workspace.get(dev)
+        if cached is None or cached[0] < n:
+            need = ops.centered_rank_sorted_workspace(n)
+            cached = (n, torch.empty(need, dtype=torch.uint8, device=dev))
+            _rank_workspace[dev] = cached
+        ws = cached[1]
+        ops.centered_rank_sorted(fitness.data_ptr(), n, out.data_ptr(),
+                                 ws.data_ptr(), ws.numel(), _stream())
+        return out
     ops.centered_rank(fitness.data_ptr(), n, out.data_ptr(), _stream())
     return out
 

@@ -90,6 +90,26 @@ class TestCenteredRank:
         want = ops.centered_rank_ref(f)
         assert torch.allclose(got, want, atol=1e-6)
 
+    def test_large_population_sorted_path(self):
+        """pop 131,072 (the named 8-GPU config) takes the rocPRIM radix
+        sort path; output must match the stable-argsort reference
+        EXACTLY, including index-order tie-breaks and negatives."""
+        torch.manual_seed(3)
+        n = 131072
+        # quantized values force heavy ties
+        f = (torch.randn(n, device="cuda") * 4).round().contiguous()
+        got = ops.centered_rank(f)
+        want = ops.centered_rank_ref(f)
+        assert torch.equal(got, want)
+
+    def test_sorted_path_boundary(self):
+        """Just past the n^2/sort switch (16384): both paths agree."""
+        torch.manual_seed(4)
+        f = torch.randn(16385, device="cuda").contiguous()
+        got = ops.centered_rank(f)
+        want = ops.centered_rank_ref(f)
+        assert torch.equal(got, want)
+
 
 @requires_gpu
 class TestPhiloxDevice:
