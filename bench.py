@@ -55,6 +55,9 @@ def main():
         from fiber_amd.ring import RingContext
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        if os.environ.get("FAM_NCCL_DEBUG"):
+            # RCCL-level diagnostics for ring/xGMI setup issues
+            os.environ.setdefault("NCCL_DEBUG", "WARN")
         ctx = RingContext(rank, world, backend="nccl", device=device)
         ctx.init()
 

@@ -127,7 +127,7 @@ class ConvESEngine:
         # of the whole rollout stays valid across ES iterations
         self._iter_buf = torch.zeros(1, dtype=torch.int32, device=device)
         self._graph = None
-        self.use_graph = True
+        self.use_graph = device.type == "cuda"
         import os as _os
 
         # 8 population chunks measured best once all MFMA layers were
@@ -136,17 +136,20 @@ class ConvESEngine:
         nhalves = int(_os.environ.get("FAM_CONV_STREAMS", "8"))
         if pop % nhalves or pop < 4 * nhalves:
             nhalves = 1
+        # CPU instantiation (distributed choreography tests with a
+        # stubbed rollout) has no HIP streams.
         self._half_streams = [
             torch.cuda.Stream(device=device) for _ in range(nhalves)
-        ]
+        ] if device.type == "cuda" else []
         # shared obs-noise staging, one e4m3 field per stream chunk: the
         # noise is keyed by (env, pos, t) only (common random numbers
         # across the population), so conv_noisegen fills this once per
         # (chunk, t) and conv_layer1 re-reads it for every member —
         # 1/(pop/chunk) of the philox work the fused kernel redid.
         # Chunks run at skewed timesteps, hence one buffer each.
-        self.znoise = torch.empty(nhalves, E * 84 * 84 * 4,
-                                  dtype=torch.uint8, device=device)
+        self.znoise = torch.empty(
+            max(1, len(self._half_streams)), E * 84 * 84 * 4,
+            dtype=torch.uint8, device=device)
 
     def _stream(self):
         return torch.cuda.current_stream().cuda_stream

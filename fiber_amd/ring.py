@@ -75,14 +75,39 @@ class RingContext:
     def init(self):
         if self._initialized:
             return self
+        import datetime
+
         import torch.distributed as dist
 
         if not dist.is_initialized():
-            dist.init_process_group(
-                backend=self.backend,
-                rank=self.rank,
-                world_size=self.size,
+            # Bounded rendezvous: a missing/dead rank surfaces as a
+            # timeout with diagnosis instead of an indefinite hang
+            # (FAM_PG_TIMEOUT seconds, default 600).
+            timeout = datetime.timedelta(
+                seconds=float(os.environ.get("FAM_PG_TIMEOUT", "600"))
             )
+            try:
+                dist.init_process_group(
+                    backend=self.backend,
+                    rank=self.rank,
+                    world_size=self.size,
+                    timeout=timeout,
+                )
+            except Exception:
+                import sys
+
+                print(
+                    "[fiber_amd.ring] init_process_group failed: "
+                    "backend=%s rank=%d world=%d MASTER=%s:%s "
+                    "HIP_VISIBLE_DEVICES=%r (set NCCL_DEBUG=WARN or "
+                    "FAM_NCCL_DEBUG=1 for RCCL-level diagnostics)"
+                    % (self.backend, self.rank, self.size,
+                       os.environ.get("MASTER_ADDR"),
+                       os.environ.get("MASTER_PORT"),
+                       os.environ.get("HIP_VISIBLE_DEVICES")),
+                    file=sys.stderr, flush=True,
+                )
+                raise
         self._initialized = True
         _ = self.device
         return self

@@ -72,6 +72,44 @@ def _rank_main(rank, size, out=None):
     ctx.shutdown()
 
 
+def _conv_rank_main(rank, size, out=None):
+    from fiber_amd import ops
+    from fiber_amd.es.conv_policy import ConvESConfig, ConvESEngine
+
+    _cpu_stub_ops(ops)
+    ctx = RingContext(rank, size, backend="gloo",
+                      device=torch.device("cpu"))
+    ctx.init()
+    cfg = ConvESConfig(pop_per_gpu=4, horizon=2, seed=77)
+    engine = ConvESEngine(cfg, ctx=ctx, device=torch.device("cpu"))
+
+    # CPU stand-in for the HIP rollout pipeline: fitness depends only on
+    # the GLOBAL member id (so gather ordering is checkable) + theta.
+    def rollout(iteration):
+        member = torch.arange(rank * cfg.pop_per_gpu,
+                              (rank + 1) * cfg.pop_per_gpu,
+                              dtype=torch.float32)
+        return torch.cos(member * 0.9) + 0.001 * float(engine.theta.sum())
+
+    engine.rollout = rollout
+    # the conv param space is huge; cap the stub grad work
+    import fiber_amd.es.conv_policy as cp
+
+    def es_grad_small(wpair, pair_begin, pair_end, seed, iteration, device,
+                      nparams=None):
+        grad = torch.zeros(nparams)
+        for pair in range(pair_begin, pair_end):
+            grad[pair % nparams] += float(wpair[pair])
+        return grad
+
+    cp.ops.es_grad = es_grad_small
+    for _ in range(3):
+        stats = engine.step()
+    out.put((rank, stats["fitness_mean"],
+             engine.theta.double().sum().item()))
+    ctx.shutdown()
+
+
 class TestDistributedES:
     def test_two_rank_step_theta_identical(self):
         out = SimpleQueue()
@@ -81,11 +119,26 @@ class TestDistributedES:
         results = sorted(out.get(timeout=10) for _ in range(2))
         (r0, fit0, sum0, head0), (r1, fit1, sum1, head1) = results
         assert (r0, r1) == (0, 1)
-        # both ranks saw the same全 population fitness
+        # both ranks saw the same full-population fitness
         assert fit0 == fit1
         # theta stays bit-identical across ranks after 3 steps
         assert sum0 == sum1
         assert head0 == head1
+        out.close()
+
+    def test_conv_two_rank_step_theta_identical(self):
+        """Conv-engine N>1 choreography (VERDICT next-round #3): the
+        same fitness all-gather + grad all-reduce + identical-Adam
+        invariants as the MLP engine, on CPU/gloo with world 2."""
+        out = SimpleQueue()
+        ring = Ring(2, functools.partial(_conv_rank_main, out=out),
+                    backend="gloo", gpu_per_rank=0)
+        ring.run(timeout=300)
+        results = sorted(out.get(timeout=10) for _ in range(2))
+        (r0, fit0, sum0), (r1, fit1, sum1) = results
+        assert (r0, r1) == (0, 1)
+        assert fit0 == fit1
+        assert sum0 == sum1
         out.close()
 
     def test_fitness_gather_ordering(self):
