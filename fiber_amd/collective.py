@@ -1,0 +1,227 @@
+"""Pool-owned RCCL collective plane.
+
+The north-star data path (SURVEY §2b "MI355X equivalents"): ``Pool.map``
+fan-out/fan-in rides RCCL broadcast / reduce over xGMI for tensor
+payloads, while scalar metadata keeps riding the shm rings.  This module
+provides the communicator machinery:
+
+* :class:`GroupMaster` — master-side: owns the rendezvous (a TCPStore
+  hosted in the pool master, which is alive for the pool's whole life)
+  and a **generation** counter.  A worker death invalidates the
+  communicator; the master bumps the generation, rotates the port, and
+  broadcasts a rebuild — the next collective re-initializes cleanly.
+* :class:`WorkerGroup` — worker-side: lazily joins the process group for
+  the current generation (backend ``"nccl"`` == RCCL over xGMI on GPU
+  workers, ``"gloo"`` for CPU plumbing tests) and exposes the same
+  collective surface as :class:`fiber_amd.ring.RingContext` so ES engines
+  run unchanged inside pool workers.
+
+The reference has no analog — fiber's nanomsg PUSH/PULL moved pickled
+bytes only (``/root/reference/fiber/pool.py:906-920``); tensor-aware
+fan-out/fan-in is the MI355X-native upgrade BASELINE.json names.
+"""
+
+import datetime
+import os
+import socket as _socket
+
+
+def _free_tcp_port():
+    s = _socket.socket(_socket.AF_INET, _socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _pg_timeout():
+    return datetime.timedelta(
+        seconds=float(os.environ.get("FAM_PG_TIMEOUT", "120"))
+    )
+
+
+class GroupMaster:
+    """Master-side communicator bookkeeping (no rank of its own).
+
+    The master never joins the process group — coordination rides the shm
+    rings — so a CPU-only master can drive an all-GPU worker group."""
+
+    def __init__(self, world, backend=None):
+        self.world = world
+        self.backend = backend  # None: workers decide (nccl on GPU)
+        self.gen = 0
+        self.host = "127.0.0.1"
+        self.port = None
+        self._store = None
+
+    def ensure(self):
+        if self._store is None:
+            self._new_store()
+        return self
+
+    def _new_store(self):
+        import torch.distributed as dist
+
+        self.port = _free_tcp_port()
+        # wait_for_workers=False: the server thread must not block the
+        # master; workers rendezvous against it with their own timeout.
+        self._store = dist.TCPStore(
+            self.host, self.port, self.world, True,
+            wait_for_workers=False,
+        )
+
+    def rebuild(self):
+        """Invalidate the current generation (a member died)."""
+        self.gen += 1
+        self._store = None  # drop the old server; port is rotated
+        self._new_store()
+        return self.descriptor()
+
+    def descriptor(self, rank=None):
+        self.ensure()
+        d = {
+            "gen": self.gen,
+            "world": self.world,
+            "host": self.host,
+            "port": self.port,
+            "backend": self.backend,
+        }
+        if rank is not None:
+            d["rank"] = rank
+        return d
+
+    def close(self):
+        self._store = None
+
+
+class WorkerGroup:
+    """Per-worker collective context (RingContext-compatible surface)."""
+
+    def __init__(self, desc):
+        self._desc = dict(desc)
+        self.rank = desc["rank"]
+        self._inited_gen = None
+        self._store = None
+        self._device = None
+
+    # -- membership --------------------------------------------------------
+    @property
+    def size(self):
+        return self._desc["world"]
+
+    world = size
+
+    @property
+    def gen(self):
+        return self._desc["gen"]
+
+    @property
+    def backend(self):
+        b = self._desc.get("backend")
+        if b is None:
+            import torch
+
+            b = "nccl" if torch.cuda.is_available() else "gloo"
+        return b
+
+    @property
+    def device(self):
+        import torch
+
+        if self._device is None:
+            if self.backend == "nccl":
+                # HIP_VISIBLE_DEVICES pins this worker to one MI355X.
+                self._device = torch.device("cuda", 0)
+                torch.cuda.set_device(self._device)
+            else:
+                self._device = torch.device("cpu")
+        return self._device
+
+    def apply_rebuild(self, desc):
+        """New generation from the master: tear down, re-init lazily."""
+        if desc["gen"] <= self._desc["gen"]:
+            return
+        self.destroy()
+        rank = self.rank
+        self._desc = dict(desc)
+        self._desc["rank"] = rank
+        self.rank = rank
+
+    def ensure(self):
+        if self._inited_gen == self._desc["gen"]:
+            return self
+        import torch.distributed as dist
+
+        self.destroy()
+        d = self._desc
+        self._store = dist.TCPStore(
+            d["host"], d["port"], d["world"], False, _pg_timeout()
+        )
+        prefixed = dist.PrefixStore("famgen%d" % d["gen"], self._store)
+        dist.init_process_group(
+            backend=self.backend,
+            store=prefixed,
+            rank=self.rank,
+            world_size=d["world"],
+            timeout=_pg_timeout(),
+        )
+        self._inited_gen = d["gen"]
+        _ = self.device
+        return self
+
+    def destroy(self):
+        import torch.distributed as dist
+
+        if self._inited_gen is not None and dist.is_initialized():
+            try:
+                dist.destroy_process_group()
+            except Exception:
+                pass
+        self._inited_gen = None
+        self._store = None
+
+    # -- collectives (RingContext-compatible) ------------------------------
+    def allreduce(self, tensor, average=False):
+        import torch.distributed as dist
+
+        self.ensure()
+        dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+        if average:
+            tensor /= self.size
+        return tensor
+
+    def broadcast(self, tensor, src=0):
+        import torch.distributed as dist
+
+        self.ensure()
+        dist.broadcast(tensor, src=src)
+        return tensor
+
+    def all_gather(self, tensor):
+        import torch
+        import torch.distributed as dist
+
+        self.ensure()
+        out = [torch.empty_like(tensor) for _ in range(self.size)]
+        dist.all_gather(out, tensor)
+        return out
+
+    def all_gather_into(self, out, tensor):
+        import torch.distributed as dist
+
+        self.ensure()
+        dist.all_gather_into_tensor(out, tensor)
+        return out
+
+    def reduce_scatter(self, out, tensor):
+        import torch.distributed as dist
+
+        self.ensure()
+        dist.reduce_scatter_tensor(out, tensor)
+        return out
+
+    def barrier(self):
+        import torch.distributed as dist
+
+        self.ensure()
+        dist.barrier()

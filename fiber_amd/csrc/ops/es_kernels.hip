@@ -418,7 +418,10 @@ extern "C" __global__ void rank_pack_keys(const float* __restrict__ f, int n,
                                           unsigned* __restrict__ vals) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  const unsigned u = __float_as_uint(f[i]);
+  unsigned u = __float_as_uint(f[i]);
+  // float compare (and torch's sort) treats -0.0 == +0.0; the bit trick
+  // would order them — canonicalize so they tie (index-order break).
+  if ((u & 0x7FFFFFFFu) == 0u) u = 0u;
   keys[i] = (u & 0x80000000u) ? ~u : (u | 0x80000000u);
   vals[i] = (unsigned)i;
 }

@@ -15,9 +15,21 @@ re-based on shared-memory rings:
   exceptions kill the worker (tasks must be idempotent) and the chunk is
   retried, so probabilistic failures still converge to a complete result.
 
-Task wire format: ``(seq, base, func_blob, args, starmap, kwds)``.
+Collective mode (``collective=True``, ZPool only) adds the RCCL data
+plane (SURVEY §2b "MI355X equivalents"; reference fan-out/fan-in:
+``fiber/pool.py:906-920``): each worker joins a pool-owned process group
+("nccl" == RCCL over xGMI on GPU workers), shared tensor map-args are
+IPC-shipped ONCE to rank 0 and broadcast device-to-device, tensor
+results can be summed with one all-reduce, and :meth:`ZPool.run_on_all`
+runs one SPMD task per worker (the ES-through-pool fan-out).  Control
+messages ride per-worker shm rings in a globally consistent order, so
+every worker executes the same collective sequence.
+
+Task wire format: ``(seq, base, func_blob, args, starmap, kwds, flags)``
+(``flags`` bit 0 = the map has a collective begin record).
 Result wire format: ``(seq, base, values, failure, ident)`` where
-``failure`` is None or ``(index, exception)``.
+``failure`` is None or ``(index, exception)``; ``base == -2`` carries a
+collective-reduce final tensor.
 """
 
 import itertools
@@ -33,6 +45,8 @@ from .transport import Socket
 
 DEFAULT_CHUNKSIZE = 32
 _SENTINEL_SEQ = -1
+_REDUCE_BASE = -2          # result record carrying a reduced tensor
+_FLAG_COLLECTIVE = 1       # task flags bit: wait for the map's begin ctl
 
 
 class _ExcInfo:
@@ -356,14 +370,160 @@ def _execute_chunk(func, args, starmap, kwds):
     return values, None
 
 
+_current_worker_group = None
+
+
+def current_worker_group():
+    """Inside a collective pool worker: the :class:`WorkerGroup` (rank,
+    size, allreduce/broadcast/all_gather/... over the pool's RCCL
+    communicator).  None outside collective workers."""
+    return _current_worker_group
+
+
+class _WorkerCollState:
+    """Worker-side collective bookkeeping: the group, per-map shared
+    tensors and reduce partials, and the ctl-ring processor.  Control
+    messages are executed strictly in ring order — the master sends them
+    to every worker in the same order under one lock, which is what
+    keeps collective calls matched across ranks."""
+
+    def __init__(self, ctl_addr, group_desc, result_sock, ident):
+        from .collective import WorkerGroup
+
+        self.sock = Socket("r", ctl_addr, bind=False)
+        self.group = WorkerGroup(group_desc)
+        self.result_sock = result_sock
+        self.ident = ident
+        self.maps = {}  # seq -> {shared, reduce, spec, partial}
+        global _current_worker_group
+        _current_worker_group = self.group
+
+    # -- ctl processing ----------------------------------------------------
+    def drain(self, timeout=0.0):
+        """Process pending ctl messages.  Returns False when the ctl ring
+        is gone (pool teardown) — the worker loop should exit."""
+        while True:
+            try:
+                msg = self.sock.recv(timeout)
+            except RuntimeError:
+                return False
+            if msg is None:
+                return True
+            timeout = 0.0
+            self._handle(serialization.loads(msg))
+
+    def wait_map(self, seq):
+        """Block until the begin record for ``seq`` has been processed
+        (it was ctl-broadcast before any of the map's chunks)."""
+        while seq not in self.maps:
+            if not self.drain(timeout=1.0):
+                raise RuntimeError("pool ctl channel closed")
+        return self.maps[seq]
+
+    def _handle(self, msg):
+        op = msg[0]
+        if op == "begin":
+            _, seq, meta, blob, reduce_mode, spec = msg
+            shared = self._stage_shared(meta, blob) if meta else {}
+            self.maps[seq] = {
+                "shared": shared,
+                "reduce": reduce_mode,
+                "spec": spec,
+                "partial": None,
+            }
+        elif op == "reduce_go":
+            _, seq = msg
+            self._finish_reduce(seq)
+        elif op == "drop":
+            self.maps.pop(msg[1], None)
+        elif op == "exec":
+            _, seq, func_blob, args, kwds = msg
+            self._exec(seq, func_blob, args, kwds)
+        elif op == "rebuild":
+            self.group.apply_rebuild(msg[1])
+
+    def _stage_shared(self, meta, blob):
+        """Stage the map's shared tensors: rank 0 materializes them from
+        the master's one-time blob (HIP IPC handles for device tensors),
+        every other rank allocates empty buffers, then each tensor is
+        broadcast over the pool communicator (xGMI on GPU workers) —
+        ONE host->device hand-off total instead of one per chunk."""
+        import torch
+
+        device = self.group.device
+        shared = {}
+        if self.group.rank == 0:
+            src = serialization.loads(blob)
+            for name in sorted(meta):
+                shared[name] = src[name].to(device).contiguous()
+        else:
+            for name in sorted(meta):
+                shape, dtype_str = meta[name]
+                dtype = getattr(torch, dtype_str.split(".")[-1])
+                shared[name] = torch.empty(shape, dtype=dtype,
+                                           device=device)
+        for name in sorted(meta):
+            self.group.broadcast(shared[name], src=0)
+        return shared
+
+    def accumulate(self, seq, values):
+        entry = self.maps.get(seq)
+        if entry is None:
+            return
+        for value in values:
+            if entry["partial"] is None:
+                entry["partial"] = value.clone()
+            else:
+                entry["partial"] += value
+
+    def _finish_reduce(self, seq):
+        import torch
+
+        entry = self.maps.pop(seq, None)
+        if entry is None:
+            return
+        partial = entry["partial"]
+        if partial is None:
+            shape, dtype_str = entry["spec"]
+            dtype = getattr(torch, dtype_str.split(".")[-1])
+            partial = torch.zeros(shape, dtype=dtype,
+                                  device=self.group.device)
+        self.group.allreduce(partial)
+        if self.group.rank == 0:
+            record = (seq, _REDUCE_BASE, [partial], None, self.ident)
+            self.result_sock.send(serialization.dumps(record), timeout=-1.0)
+
+    def _exec(self, seq, func_blob, args, kwds):
+        """SPMD exec: every worker runs the same call once; results land
+        at index == rank.  The function may use current_worker_group()
+        for its own collectives (the ES-through-pool hot path)."""
+        try:
+            func = serialization.loads(func_blob)
+            value = func(*args, **(kwds or {}))
+            record = (seq, self.group.rank, [value], None, self.ident)
+        except Exception as exc:  # noqa: BLE001
+            record = (seq, self.group.rank, [], (0, _ExcInfo(exc)),
+                      self.ident)
+        self.result_sock.send(serialization.dumps(record), timeout=-1.0)
+
+    def close(self):
+        global _current_worker_group
+        _current_worker_group = None
+        self.group.destroy()
+        self.sock.close()
+
+
 def _pool_worker_loop(
     task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix,
-    nproc=1,
+    nproc=1, ctl_addr=None, group_desc=None,
 ):
     """Worker job entry: runs ``nproc`` worker cores in this job
     (reference multi-worker-per-job, ``cpu_per_job``: forked cores share
     the shm rings; each core gets a derived ident for attribution).
     Forking happens before any socket/HIP state exists in the core."""
+    if ctl_addr is not None and nproc > 1:
+        raise ValueError("collective mode is incompatible with "
+                         "cpu_per_worker > 1 (one rank per process)")
     if nproc > 1:
         children = []
         for k in range(1, nproc):
@@ -405,7 +565,7 @@ def _pool_worker_loop(
                     pass
         return
     _pool_worker_core(task_addr, result_addr, resilient, maxtasks,
-                      init_blob, ident_prefix)
+                      init_blob, ident_prefix, ctl_addr, group_desc)
 
 
 def _set_pdeathsig():
@@ -421,14 +581,20 @@ def _set_pdeathsig():
 
 
 def _pool_worker_core(
-    task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix
+    task_addr, result_addr, resilient, maxtasks, init_blob, ident_prefix,
+    ctl_addr=None, group_desc=None,
 ):
     """Worker main loop (reference zpool_worker_core, pool.py:760-825)."""
+    result_sock = Socket("w", result_addr, bind=False)
+    coll = None
+    if ctl_addr is not None:
+        ident0 = ident_prefix or util.random_name("w")[:24]
+        coll = _WorkerCollState(ctl_addr, group_desc, result_sock, ident0)
+
     if init_blob is not None:
         initializer, initargs = serialization.loads(init_blob)
         initializer(*initargs)
 
-    result_sock = Socket("w", result_addr, bind=False)
     func_cache = {}
     tasks_done = 0
 
@@ -439,19 +605,31 @@ def _pool_worker_core(
         task_sock = Socket("r", task_addr, bind=False)
 
     while True:
+        if coll is not None and not coll.drain():
+            break  # ctl ring gone: pool tearing down
         if resilient:
             task_sock.send(b"", timeout=-1.0)
             # Timeout + re-request guards against a discarded request (the
             # master drops requests it cannot attribute to a live worker).
             payload = task_sock.recv_view(timeout=10.0)
         else:
-            payload = task_sock.recv_view(timeout=-1.0)
+            # Collective workers poll so ctl messages (staging
+            # broadcasts, SPMD execs) are served between chunks.
+            payload = task_sock.recv_view(
+                timeout=0.2 if coll is not None else -1.0
+            )
         if payload is None:
             continue
         task = serialization.loads(payload)
-        seq, base, func_blob, args, star, kwds = task
+        seq, base, func_blob, args, star, kwds, flags = task
         if seq == _SENTINEL_SEQ:
             break
+        entry = None
+        if flags & _FLAG_COLLECTIVE and coll is not None:
+            entry = coll.wait_map(seq)
+            if entry["shared"]:
+                kwds = dict(kwds or {})
+                kwds.update(entry["shared"])
         # Keyed by the blob bytes themselves: a hash collision between two
         # distinct pickled functions must not silently run the wrong one.
         key = bytes(func_blob)
@@ -467,6 +645,10 @@ def _pool_worker_core(
             # resubmit the whole chunk (tasks must be idempotent —
             # duplicate deliveries are deduped master-side).
             raise failure[1].rebuild()
+        if entry is not None and entry["reduce"] and failure is None:
+            # tensor fan-in rides the all-reduce at map end, not the ring
+            coll.accumulate(seq, values)
+            values = [None] * len(values)
         result = (seq, base, values, failure, ident)
         result_sock.send(serialization.dumps(result), timeout=-1.0)
 
@@ -474,6 +656,8 @@ def _pool_worker_core(
         if maxtasks is not None and tasks_done >= maxtasks:
             break
 
+    if coll is not None:
+        coll.close()
     task_sock.close()
     result_sock.close()
 
@@ -508,6 +692,8 @@ class ZPool:
         gpu_per_worker=None,
         cpu_per_worker=None,
         name=None,
+        collective=False,
+        collective_backend=None,
     ):
         conf = fam_config.get_object()
         self._processes = processes or os.cpu_count() or 1
@@ -518,6 +704,34 @@ class ZPool:
         self._meta = {}
         if gpu_per_worker:
             self._meta["gpu"] = gpu_per_worker
+
+        # RCCL data plane: a pool-owned communicator over
+        # the workers.  Deterministic membership is required (rank ==
+        # worker slot), so it is a ZPool feature; ResilientZPool's
+        # anonymous-requeue scheduling is incompatible by design.
+        self._collective = bool(collective)
+        self._group_master = None
+        self._ctl_socks = {}       # ident -> Socket("w") per-worker ctl
+        self._ctl_lock = threading.Lock()
+        self._slots = [None] * self._processes  # slot -> ident
+        self._coll_seqs = set()    # in-flight collective map seqs
+        self._reduce_pending = {}  # seq -> spec (reduce_go not yet sent)
+        self._reduce_results = {}  # seq -> [Event, tensor, error]
+        if self._collective:
+            if self.resilient:
+                raise ValueError(
+                    "collective=True requires ZPool: in-flight "
+                    "collectives cannot be resubmitted to a different "
+                    "rank (use ZPool and handle map failures yourself)"
+                )
+            if self._nproc_per_job > 1:
+                raise ValueError("collective=True needs one rank per "
+                                 "worker process (cpu_per_worker == 1)")
+            from .collective import GroupMaster
+
+            self._group_master = GroupMaster(
+                self._processes, collective_backend
+            )
 
         self._init_blob = None
         if initializer is not None:
@@ -570,34 +784,40 @@ class ZPool:
         )
         self._worker_thread.start()
 
-    def _spawn_worker(self, index):
+    def _spawn_worker(self, slot):
         ident = util.random_name("w")[:24]
         self._all_idents.add(ident)
-        entry = _WorkerEntry(
-            dict(
-                task_addr=self._name + ".task",
-                result_addr=self._name + ".res",
-                resilient=self.resilient,
-                maxtasks=self._maxtasks,
-                init_blob=self._init_blob,
-                ident_prefix=ident,
-                nproc=self._nproc_per_job,
-            ),
-            self._meta,
+        kwargs = dict(
+            task_addr=self._name + ".task",
+            result_addr=self._name + ".res",
+            resilient=self.resilient,
+            maxtasks=self._maxtasks,
+            init_blob=self._init_blob,
+            ident_prefix=ident,
+            nproc=self._nproc_per_job,
         )
+        if self._collective:
+            # rank == slot: a respawned worker inherits the dead one's
+            # rank, so the communicator shape is stable across deaths.
+            ctl_addr = "%s.ctl.%s" % (self._name, ident)
+            with self._ctl_lock:
+                self._ctl_socks[ident] = Socket("w", ctl_addr, bind=True)
+            kwargs["ctl_addr"] = ctl_addr
+            kwargs["group_desc"] = self._group_master.descriptor(rank=slot)
+        entry = _WorkerEntry(kwargs, self._meta)
         proc = Process(
-            target=entry, name="%s-worker-%d" % (self._name, index)
+            target=entry, name="%s-worker-%d" % (self._name, slot)
         )
         # Register BEFORE start so the dispatcher can attribute the
         # worker's very first task request.
         with self._worker_lock:
             self._workers[ident] = proc
+            self._slots[slot] = ident
         proc.start()
         return ident, proc
 
     def _worker_loop(self):
         """Maintain the worker population; resubmit a dead worker's tasks."""
-        index = itertools.count()
         while self._state == "run":
             with self._worker_lock:
                 dead = [
@@ -607,20 +827,76 @@ class ZPool:
                 ]
                 for ident, proc in dead:
                     del self._workers[ident]
+                    for slot, sid in enumerate(self._slots):
+                        if sid == ident:
+                            self._slots[slot] = None
             for ident, proc in dead:
                 self._on_worker_death(ident, proc)
+            if dead and self._collective:
+                self._rebuild_group([i for i, _ in dead])
             with self._worker_lock:
-                missing = self._processes - len(self._workers)
-            for _ in range(missing):
+                free = [s for s, sid in enumerate(self._slots)
+                        if sid is None]
+            for slot in free:
                 if self._state != "run":
                     break
                 try:
-                    self._spawn_worker(next(index))
+                    self._spawn_worker(slot)
                 except Exception:
                     if self._state == "run":
                         util.get_logger().exception("worker spawn failed")
                     time.sleep(0.5)
             time.sleep(0.1)
+
+    def _rebuild_group(self, dead_idents):
+        """A member died: the communicator generation is invalid.  Fail
+        every in-flight collective map with diagnosis (Ring-failure
+        policy: fail fast, rebuild for subsequent work), rotate the
+        rendezvous, and tell the survivors."""
+        for ident in dead_idents:
+            with self._ctl_lock:
+                sock = self._ctl_socks.pop(ident, None)
+            if sock is not None:
+                sock.close()
+        exc = RuntimeError(
+            "pool worker(s) %s died during a collective map; the "
+            "communicator was rebuilt — resubmit the map" % dead_idents
+        )
+        for seq in list(self._coll_seqs):
+            self._coll_seqs.discard(seq)
+            self._reduce_pending.pop(seq, None)
+            holder = self._reduce_results.get(seq)
+            if holder is not None:
+                holder[2] = exc
+                holder[0].set()
+            self._inventory.fail_seq(seq, exc)
+        desc = self._group_master.rebuild()
+        self._ctl_broadcast(lambda rank, ident: ("rebuild", desc))
+        util.get_logger().warning(
+            "collective group rebuilt (gen %d) after death of %s",
+            desc["gen"], dead_idents,
+        )
+
+    def _ctl_broadcast(self, make_msg):
+        """Send a ctl record to every live worker under ONE lock hold —
+        the global ordering guarantee the worker-side collective
+        sequencing relies on.  make_msg(rank, ident) -> tuple."""
+        with self._ctl_lock:
+            with self._worker_lock:
+                slots = list(enumerate(self._slots))
+            for rank, ident in slots:
+                if ident is None:
+                    continue
+                sock = self._ctl_socks.get(ident)
+                if sock is None:
+                    continue
+                try:
+                    sock.send(
+                        serialization.dumps(make_msg(rank, ident)),
+                        timeout=-1.0,
+                    )
+                except (RuntimeError, OSError):
+                    pass  # dying worker; the reaper handles it
 
     def _on_worker_death(self, ident, proc):
         pass  # resilient subclass resubmits
