@@ -24,6 +24,7 @@ fan-out/fan-in is the MI355X-native upgrade BASELINE.json names.
 import datetime
 import os
 import socket as _socket
+import threading
 
 
 def _free_tcp_port():
@@ -40,6 +41,13 @@ def _pg_timeout():
     )
 
 
+class CollectiveError(RuntimeError):
+    """A communicator-level failure (rendezvous / transport), as opposed
+    to a user-code exception.  The pool master reacts by rotating the
+    group generation so retries start from a virgin namespace instead of
+    re-using half-initialized gloo/RCCL state."""
+
+
 class GroupMaster:
     """Master-side communicator bookkeeping (no rank of its own).
 
@@ -53,10 +61,15 @@ class GroupMaster:
         self.host = "127.0.0.1"
         self.port = None
         self._store = None
+        # ensure() races between the worker-spawn thread (descriptor())
+        # and the submitting thread; a double-create would orphan the
+        # store half the workers were told about.
+        self._lock = threading.Lock()
 
     def ensure(self):
-        if self._store is None:
-            self._new_store()
+        with self._lock:
+            if self._store is None:
+                self._new_store()
         return self
 
     def _new_store(self):
@@ -72,9 +85,10 @@ class GroupMaster:
 
     def rebuild(self):
         """Invalidate the current generation (a member died)."""
-        self.gen += 1
-        self._store = None  # drop the old server; port is rotated
-        self._new_store()
+        with self._lock:
+            self.gen += 1
+            self._store = None  # drop the old server; port is rotated
+            self._new_store()
         return self.descriptor()
 
     def descriptor(self, rank=None):
@@ -158,16 +172,45 @@ class WorkerGroup:
             d["host"], d["port"], d["world"], False, _pg_timeout()
         )
         prefixed = dist.PrefixStore("famgen%d" % d["gen"], self._store)
-        dist.init_process_group(
-            backend=self.backend,
-            store=prefixed,
-            rank=self.rank,
-            world_size=d["world"],
-            timeout=_pg_timeout(),
-        )
+        # init_process_group namespaces its store keys with a
+        # PROCESS-LOCAL group counter; after failed attempts the counter
+        # skews across ranks and they rendezvous on different keys
+        # forever.  This process owns only the pool group, so pin the
+        # counter: every rank of a generation meets at /famgenG/0/.
+        try:
+            from torch.distributed.distributed_c10d import _world
+
+            _world.group_count = 0
+        except Exception:  # pragma: no cover - private API moved
+            pass
+        try:
+            dist.init_process_group(
+                backend=self.backend,
+                store=prefixed,
+                rank=self.rank,
+                world_size=d["world"],
+                timeout=_pg_timeout(),
+            )
+        except Exception as exc:
+            self.destroy()
+            raise CollectiveError(
+                "group init failed (gen %d): %s" % (d["gen"], exc)
+            ) from exc
         self._inited_gen = d["gen"]
         _ = self.device
         return self
+
+    def _collective(self, op):
+        """Run one collective; a transport failure resets local state and
+        surfaces as CollectiveError (master rotates the generation)."""
+        self.ensure()
+        try:
+            return op()
+        except Exception as exc:
+            self.destroy()
+            raise CollectiveError(
+                "collective failed (gen %d): %s" % (self._desc["gen"], exc)
+            ) from exc
 
     def destroy(self):
         import torch.distributed as dist
@@ -184,44 +227,47 @@ class WorkerGroup:
     def allreduce(self, tensor, average=False):
         import torch.distributed as dist
 
-        self.ensure()
-        dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
-        if average:
-            tensor /= self.size
-        return tensor
+        def op():
+            dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+            if average:
+                tensor.div_(self.size)
+            return tensor
+
+        return self._collective(op)
 
     def broadcast(self, tensor, src=0):
         import torch.distributed as dist
 
-        self.ensure()
-        dist.broadcast(tensor, src=src)
-        return tensor
+        return self._collective(
+            lambda: (dist.broadcast(tensor, src=src), tensor)[1]
+        )
 
     def all_gather(self, tensor):
         import torch
         import torch.distributed as dist
 
-        self.ensure()
-        out = [torch.empty_like(tensor) for _ in range(self.size)]
-        dist.all_gather(out, tensor)
-        return out
+        def op():
+            out = [torch.empty_like(tensor) for _ in range(self.size)]
+            dist.all_gather(out, tensor)
+            return out
+
+        return self._collective(op)
 
     def all_gather_into(self, out, tensor):
         import torch.distributed as dist
 
-        self.ensure()
-        dist.all_gather_into_tensor(out, tensor)
-        return out
+        return self._collective(
+            lambda: (dist.all_gather_into_tensor(out, tensor), out)[1]
+        )
 
     def reduce_scatter(self, out, tensor):
         import torch.distributed as dist
 
-        self.ensure()
-        dist.reduce_scatter_tensor(out, tensor)
-        return out
+        return self._collective(
+            lambda: (dist.reduce_scatter_tensor(out, tensor), out)[1]
+        )
 
     def barrier(self):
         import torch.distributed as dist
 
-        self.ensure()
-        dist.barrier()
+        return self._collective(lambda: dist.barrier())

@@ -205,6 +205,20 @@ class Inventory:
                     job["results"][index] = exc_info
             self._cond.notify_all()
 
+    def fail_seq(self, seq, exc):
+        """Abort one job (e.g. a collective map whose group died)."""
+        with self._cond:
+            job = self._jobs.get(seq)
+            if job is None:
+                return
+            if job.get("dyn"):
+                if job["error"] is None:
+                    job["error"] = _ExcInfo(exc)
+            elif job["remaining"] > 0:
+                job["error"] = _ExcInfo(exc)
+                job["remaining"] = 0
+            self._cond.notify_all()
+
     def fail_all(self, exc):
         """Abort every outstanding job (pool terminated)."""
         with self._cond:
@@ -352,6 +366,40 @@ MapResult = AsyncResult
 ApplyResult = AsyncResult
 
 
+class ReduceResult:
+    """Result of ``map(..., reduce='sum')``: ONE tensor — the sum of
+    every task's return value — produced worker-side by the pool-wide
+    RCCL all-reduce (fan-in over xGMI), shipped to the master once."""
+
+    def __init__(self, pool, seq):
+        self._pool = pool
+        self._seq = seq
+
+    def get(self, timeout=None):
+        holder = self._pool._reduce_results.get(self._seq)
+        if holder is None:
+            raise RuntimeError("reduce result already collected")
+        if not holder[0].wait(timeout):
+            raise TimeoutError("reduce result timed out")
+        self._pool._reduce_results.pop(self._seq, None)
+        try:  # free the placeholder inventory slot
+            self._pool._inventory.get(self._seq, 0.001)
+        except Exception:
+            pass
+        if holder[2] is not None:
+            raise holder[2]
+        return holder[1]
+
+    def wait(self, timeout=None):
+        holder = self._pool._reduce_results.get(self._seq)
+        if holder is not None:
+            holder[0].wait(timeout)
+
+    def ready(self):
+        holder = self._pool._reduce_results.get(self._seq)
+        return holder is None or holder[0].is_set()
+
+
 # ---------------------------------------------------------------------------
 # Worker side
 # ---------------------------------------------------------------------------
@@ -363,6 +411,8 @@ def _execute_chunk(func, args, starmap, kwds):
         try:
             if starmap:
                 values.append(func(*arg, **(kwds or {})))
+            elif kwds:  # shared tensors arrive as keyword args
+                values.append(func(arg, **kwds))
             else:
                 values.append(func(arg))
         except Exception as exc:  # noqa: BLE001
@@ -391,7 +441,11 @@ class _WorkerCollState:
         from .collective import WorkerGroup
 
         self.sock = Socket("r", ctl_addr, bind=False)
-        self.group = WorkerGroup(group_desc)
+        desc = dict(group_desc)
+        # Seqs of collective maps that failed before this worker spawned
+        # (their orphan chunks may still sit in the task ring).
+        self.dropped = set(desc.pop("dropped_seqs", ()))
+        self.group = WorkerGroup(desc)
         self.result_sock = result_sock
         self.ident = ident
         self.maps = {}  # seq -> {shared, reduce, spec, partial}
@@ -401,23 +455,77 @@ class _WorkerCollState:
     # -- ctl processing ----------------------------------------------------
     def drain(self, timeout=0.0):
         """Process pending ctl messages.  Returns False when the ctl ring
-        is gone (pool teardown) — the worker loop should exit."""
+        is gone (pool teardown) — the worker loop should exit.
+
+        The whole backlog is swept first so that control-plane records
+        (rebuild / drop — monotonic and collective-free) apply BEFORE any
+        queued collective op: a `begin` for a map that failed while this
+        worker was booting must be skipped, not staged against a dead
+        communicator generation (that was a live cascade: stage ->
+        connect to a dropped store -> worker death -> another rebuild)."""
+        msgs = []
         while True:
             try:
-                msg = self.sock.recv(timeout)
+                msg = self.sock.recv(timeout if not msgs else 0.0)
             except RuntimeError:
                 return False
             if msg is None:
-                return True
+                break
+            msgs.append(serialization.loads(msg))
             timeout = 0.0
-            self._handle(serialization.loads(msg))
+        pending = []
+        for msg in msgs:
+            if msg[0] == "rebuild":
+                self.group.apply_rebuild(msg[1])
+            elif msg[0] == "drop":
+                self.maps.pop(msg[1], None)
+                self.dropped.add(msg[1])
+            else:
+                pending.append(msg)
+        for msg in pending:  # collective ops keep their FIFO order
+            if msg[0] in ("begin", "reduce_go") and msg[1] in self.dropped:
+                continue
+            try:
+                self._handle(msg)
+            except Exception as exc:  # noqa: BLE001
+                # A failed collective must not kill the worker: mark the
+                # map dead locally and tell the master (usually the
+                # master already failed it via the rebuild path).
+                util.get_logger().warning(
+                    "collective ctl op %r failed: %r", msg[0], exc
+                )
+                seq = msg[1] if len(msg) > 1 else None
+                if seq is not None:
+                    self.maps.pop(seq, None)
+                    self.dropped.add(seq)
+                    record = (seq, _REDUCE_BASE, [],
+                              (self.group.gen, _ExcInfo(exc)), self.ident)
+                    try:
+                        self.result_sock.send(
+                            serialization.dumps(record), timeout=1.0
+                        )
+                    except Exception:  # noqa: BLE001
+                        pass
+        return True
 
     def wait_map(self, seq):
         """Block until the begin record for ``seq`` has been processed
-        (it was ctl-broadcast before any of the map's chunks)."""
+        (it was ctl-broadcast before any of the map's chunks).  Returns
+        None for a dropped map (failed collective) — the chunk is an
+        orphan and must be skipped."""
+        deadline = time.monotonic() + float(
+            os.environ.get("FAM_CTL_WAIT", "60")
+        )
         while seq not in self.maps:
+            if seq in self.dropped:
+                return None
             if not self.drain(timeout=1.0):
                 raise RuntimeError("pool ctl channel closed")
+            if time.monotonic() > deadline:
+                # self-healing fallback: treat as dropped rather than
+                # wedging the worker forever
+                self.dropped.add(seq)
+                return None
         return self.maps[seq]
 
     def _handle(self, msg):
@@ -434,13 +542,10 @@ class _WorkerCollState:
         elif op == "reduce_go":
             _, seq = msg
             self._finish_reduce(seq)
-        elif op == "drop":
-            self.maps.pop(msg[1], None)
         elif op == "exec":
             _, seq, func_blob, args, kwds = msg
             self._exec(seq, func_blob, args, kwds)
-        elif op == "rebuild":
-            self.group.apply_rebuild(msg[1])
+        # "rebuild"/"drop" are consumed by drain() before this point
 
     def _stage_shared(self, meta, blob):
         """Stage the map's shared tensors: rank 0 materializes them from
@@ -497,10 +602,17 @@ class _WorkerCollState:
         """SPMD exec: every worker runs the same call once; results land
         at index == rank.  The function may use current_worker_group()
         for its own collectives (the ES-through-pool hot path)."""
+        from .collective import CollectiveError
+
         try:
             func = serialization.loads(func_blob)
             value = func(*args, **(kwds or {}))
             record = (seq, self.group.rank, [value], None, self.ident)
+        except CollectiveError as exc:
+            # communicator fault, not user code: report with the
+            # generation so the master rotates the group
+            record = (seq, _REDUCE_BASE, [],
+                      (self.group.gen, _ExcInfo(exc)), self.ident)
         except Exception as exc:  # noqa: BLE001
             record = (seq, self.group.rank, [], (0, _ExcInfo(exc)),
                       self.ident)
@@ -614,9 +726,11 @@ def _pool_worker_core(
             payload = task_sock.recv_view(timeout=10.0)
         else:
             # Collective workers poll so ctl messages (staging
-            # broadcasts, SPMD execs) are served between chunks.
+            # broadcasts, SPMD execs) are served between chunks; 5 ms
+            # bounds the ctl latency at ~200 idle wakeups/s.  (SPMD
+            # callers amortize further by batching iterations per exec.)
             payload = task_sock.recv_view(
-                timeout=0.2 if coll is not None else -1.0
+                timeout=0.005 if coll is not None else -1.0
             )
         if payload is None:
             continue
@@ -627,6 +741,12 @@ def _pool_worker_core(
         entry = None
         if flags & _FLAG_COLLECTIVE and coll is not None:
             entry = coll.wait_map(seq)
+            if entry is None:
+                # orphan chunk of a failed collective map: acknowledge
+                # (flow-control) but do not execute
+                record = (seq, base, [None] * len(args), None, ident)
+                result_sock.send(serialization.dumps(record), timeout=-1.0)
+                continue
             if entry["shared"]:
                 kwds = dict(kwds or {})
                 kwds.update(entry["shared"])
@@ -715,6 +835,7 @@ class ZPool:
         self._ctl_lock = threading.Lock()
         self._slots = [None] * self._processes  # slot -> ident
         self._coll_seqs = set()    # in-flight collective map seqs
+        self._coll_dropped = set()  # failed seqs (orphan chunks linger)
         self._reduce_pending = {}  # seq -> spec (reduce_go not yet sent)
         self._reduce_results = {}  # seq -> [Event, tensor, error]
         if self._collective:
@@ -803,7 +924,9 @@ class ZPool:
             with self._ctl_lock:
                 self._ctl_socks[ident] = Socket("w", ctl_addr, bind=True)
             kwargs["ctl_addr"] = ctl_addr
-            kwargs["group_desc"] = self._group_master.descriptor(rank=slot)
+            desc = self._group_master.descriptor(rank=slot)
+            desc["dropped_seqs"] = sorted(self._coll_dropped)
+            kwargs["group_desc"] = desc
         entry = _WorkerEntry(kwargs, self._meta)
         proc = Process(
             target=entry, name="%s-worker-%d" % (self._name, slot)
@@ -862,8 +985,10 @@ class ZPool:
             "pool worker(s) %s died during a collective map; the "
             "communicator was rebuilt — resubmit the map" % dead_idents
         )
-        for seq in list(self._coll_seqs):
+        failed = list(self._coll_seqs)
+        for seq in failed:
             self._coll_seqs.discard(seq)
+            self._coll_dropped.add(seq)
             self._reduce_pending.pop(seq, None)
             holder = self._reduce_results.get(seq)
             if holder is not None:
@@ -872,6 +997,8 @@ class ZPool:
             self._inventory.fail_seq(seq, exc)
         desc = self._group_master.rebuild()
         self._ctl_broadcast(lambda rank, ident: ("rebuild", desc))
+        for seq in failed:  # orphan chunks must not wedge survivors
+            self._ctl_broadcast(lambda rank, ident: ("drop", seq))
         util.get_logger().warning(
             "collective group rebuilt (gen %d) after death of %s",
             desc["gen"], dead_idents,
@@ -964,8 +1091,52 @@ class ZPool:
                     serialization.loads(payload)
                 )
                 self._recv += 1
+                if base == _REDUCE_BASE:
+                    # reduced tensor from rank 0, or a worker-side
+                    # collective failure report
+                    if failure is not None:
+                        self._coll_seqs.discard(seq)
+                        self._reduce_pending.pop(seq, None)
+                        exc = failure[1].rebuild()
+                        holder = self._reduce_results.get(seq)
+                        if holder is not None and not holder[0].is_set():
+                            holder[2] = exc
+                            holder[0].set()
+                        self._inventory.fail_seq(seq, exc)
+                        # a communicator fault of the CURRENT generation
+                        # poisons gloo/RCCL state on every rank: rotate
+                        # to a virgin namespace so retries converge
+                        if failure[0] == self._group_master.gen:
+                            self._rebuild_group([])
+                        continue
+                    holder = self._reduce_results.get(seq)
+                    if holder is not None:
+                        holder[1] = values[0]
+                        holder[0].set()
+                    self._coll_seqs.discard(seq)
+                    continue
                 self._ack(ident, seq, base)
                 self._inventory.put(seq, base, values, failure)
+                if seq in self._reduce_pending and self._inventory.done(seq):
+                    spec = self._reduce_pending.pop(seq)
+                    done, error, _results = self._inventory.peek(seq)
+                    if error is not None:
+                        holder = self._reduce_results.get(seq)
+                        if holder is not None:
+                            holder[2] = error.rebuild()
+                            holder[0].set()
+                        self._coll_seqs.discard(seq)
+                        self._ctl_broadcast(
+                            lambda rank, ident: ("drop", seq)
+                        )
+                    else:
+                        # every chunk accumulated worker-side: fire the
+                        # all-reduce fan-in
+                        self._ctl_broadcast(
+                            lambda rank, ident: ("reduce_go", seq)
+                        )
+                elif seq in self._coll_seqs and self._inventory.done(seq):
+                    self._coll_seqs.discard(seq)
                 watcher = self._callback_watch.get(seq)
                 if watcher is not None:
                     done, error, results = self._inventory.peek(seq)
@@ -982,16 +1153,37 @@ class ZPool:
             raise ValueError("Pool not running")
 
     def _submit(self, func, iterable, chunksize, starmap, kwds=None,
-                single=False, callback=None, error_callback=None):
+                single=False, callback=None, error_callback=None,
+                shared=None, reduce=None, reduce_spec=None):
         self._check_running()
         self._lazy_start_workers(func)
+        flags = 0
+        if shared is not None or reduce is not None:
+            if not self._collective:
+                raise ValueError(
+                    "shared=/reduce= need a Pool(collective=True)"
+                )
+            if reduce is not None and reduce != "sum":
+                raise ValueError("reduce must be 'sum' (or None)")
+            if reduce is not None and reduce_spec is None:
+                raise ValueError(
+                    "reduce='sum' needs reduce_spec=(shape, dtype): a "
+                    "worker that drew no chunk must still contribute "
+                    "zeros of the right shape to the all-reduce"
+                )
+            flags |= _FLAG_COLLECTIVE
         items = list(iterable)
         n = len(items)
         seq = self._inventory.add(n)
-        result = AsyncResult(
-            self, seq, n, callback=callback, error_callback=error_callback,
-            single=single,
-        )
+        if reduce is not None:
+            result = ReduceResult(self, seq)
+        else:
+            result = AsyncResult(
+                self, seq, n, callback=callback,
+                error_callback=error_callback, single=single,
+            )
+        if flags:
+            self._begin_collective(seq, shared, reduce, reduce_spec)
         if n == 0:
             self._inventory.put(seq, 0, [], None)
             if callback is not None:
@@ -1004,8 +1196,79 @@ class ZPool:
             chunksize = max(1, min(DEFAULT_CHUNKSIZE, n // 4 or 1))
         for base in range(0, n, chunksize):
             chunk = items[base : base + chunksize]
-            self._taskq.put((seq, base, func_blob, chunk, starmap, kwds))
+            self._taskq.put(
+                (seq, base, func_blob, chunk, starmap, kwds, flags)
+            )
         return result
+
+    def _wait_all_slots(self, timeout=60.0):
+        """Block until every worker slot is spawned (its ctl ring exists
+        master-side, so broadcasts sent from now on cannot be lost even
+        if the worker process is still booting)."""
+        deadline = time.monotonic() + timeout
+        while True:
+            with self._worker_lock:
+                if all(s is not None for s in self._slots):
+                    return
+            if self._state != "run" or time.monotonic() > deadline:
+                raise TimeoutError("pool workers failed to spawn")
+            time.sleep(0.01)
+
+    def _begin_collective(self, seq, shared, reduce, reduce_spec):
+        """Stage a collective map: ONE blob of the shared tensors (HIP
+        IPC handles for device tensors) goes to rank 0; everyone else
+        gets shapes/dtypes; the device-to-device broadcast happens
+        worker-side over the pool communicator."""
+        self._group_master.ensure()
+        self._wait_all_slots()
+        meta = None
+        blob = None
+        if shared:
+            meta = {
+                name: (tuple(t.shape), str(t.dtype))
+                for name, t in shared.items()
+            }
+            blob = serialization.dumps(dict(shared))
+        spec = None
+        if reduce is not None:
+            shape, dtype = reduce_spec
+            spec = (tuple(shape), str(dtype))
+            self._reduce_pending[seq] = spec
+            self._reduce_results[seq] = [threading.Event(), None, None]
+        self._coll_seqs.add(seq)
+        self._ctl_broadcast(
+            lambda rank, ident: (
+                "begin", seq, meta, blob if rank == 0 else None,
+                reduce, spec,
+            )
+        )
+
+    def run_on_all(self, func, args=(), kwds=None, timeout=None):
+        """SPMD fan-out: run ``func(*args)`` once on EVERY worker (by
+        rank), collect one result per rank, ordered by rank.  The
+        function runs with :func:`current_worker_group` available, so it
+        can drive its own RCCL collectives — this is the ES-through-pool
+        hot path (reference analog: pool.map of rollout shards,
+        /root/reference/examples/gecco-2020/es.py:17-34, upgraded to
+        SPMD + xGMI collectives)."""
+        self._check_running()
+        if not self._collective:
+            raise ValueError("run_on_all needs a Pool(collective=True)")
+        self._lazy_start_workers(func)
+        self._group_master.ensure()
+        self._wait_all_slots()
+        seq = self._inventory.add(self._processes)
+        self._coll_seqs.add(seq)
+        func_blob = serialization.dumps_closure(func)
+        self._ctl_broadcast(
+            lambda rank, ident: ("exec", seq, func_blob, tuple(args),
+                                 kwds)
+        )
+        result = AsyncResult(self, seq, self._processes)
+        try:
+            return result.get(timeout)
+        finally:
+            self._coll_seqs.discard(seq)
 
     def stats(self):
         """Observability counters (reference only had sent/recv flow
@@ -1035,24 +1298,34 @@ class ZPool:
             callback=callback, error_callback=error_callback,
         )
 
-    def map(self, func, iterable, chunksize=None):
-        return self.map_async(func, iterable, chunksize).get()
+    def map(self, func, iterable, chunksize=None, shared=None, reduce=None,
+            reduce_spec=None):
+        return self.map_async(func, iterable, chunksize, shared=shared,
+                              reduce=reduce, reduce_spec=reduce_spec).get()
 
     def map_async(self, func, iterable, chunksize=None, callback=None,
-                  error_callback=None):
+                  error_callback=None, shared=None, reduce=None,
+                  reduce_spec=None):
         return self._submit(
             func, iterable, chunksize, False, callback=callback,
-            error_callback=error_callback,
+            error_callback=error_callback, shared=shared, reduce=reduce,
+            reduce_spec=reduce_spec,
         )
 
-    def starmap(self, func, iterable, chunksize=None):
-        return self.starmap_async(func, iterable, chunksize).get()
+    def starmap(self, func, iterable, chunksize=None, shared=None,
+                reduce=None, reduce_spec=None):
+        return self.starmap_async(
+            func, iterable, chunksize, shared=shared, reduce=reduce,
+            reduce_spec=reduce_spec,
+        ).get()
 
     def starmap_async(self, func, iterable, chunksize=None, callback=None,
-                      error_callback=None):
+                      error_callback=None, shared=None, reduce=None,
+                      reduce_spec=None):
         return self._submit(
             func, iterable, chunksize, True, callback=callback,
-            error_callback=error_callback,
+            error_callback=error_callback, shared=shared, reduce=reduce,
+            reduce_spec=reduce_spec,
         )
 
     def imap(self, func, iterable, chunksize=1):
@@ -1089,7 +1362,7 @@ class ZPool:
                         self._inventory.finish_dynamic(seq, base)
                         return
                     self._taskq.put(
-                        (seq, base, func_blob, chunk, False, None)
+                        (seq, base, func_blob, chunk, False, None, 0)
                     )
                     base += len(chunk)
             except Exception as exc:  # noqa: BLE001 (iterator may raise)
@@ -1108,7 +1381,7 @@ class ZPool:
         # One exit sentinel per worker core rides the normal task channel
         # (via the local queue) so it lands after all real tasks.
         for _ in range(self._processes * self._nproc_per_job):
-            self._taskq.put((_SENTINEL_SEQ, 0, b"", [], False, None))
+            self._taskq.put((_SENTINEL_SEQ, 0, b"", [], False, None, 0))
 
     def terminate(self):
         if self._state == "terminated":
@@ -1162,6 +1435,17 @@ class ZPool:
             self._worker_thread.join(timeout=2.0)
         self._task_sock.close()
         self._result_sock.close()
+        with self._ctl_lock:
+            for sock in self._ctl_socks.values():
+                sock.close()
+            self._ctl_socks.clear()
+        if self._group_master is not None:
+            self._group_master.close()
+        exc = RuntimeError("pool closed before reduce completed")
+        for holder in self._reduce_results.values():
+            if not holder[0].is_set():
+                holder[2] = exc
+                holder[0].set()
         # Workers killed by SIGTERM/SIGKILL never ran their atexit
         # cleanup; reap their per-ident reply rings (resilient mode)
         # master-side, including forked cores' derived idents.

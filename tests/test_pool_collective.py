@@ -1,0 +1,130 @@
+"""Pool RCCL data-plane tests (CPU plumbing: gloo, world 2).
+
+The same code path carries RCCL over xGMI on GPU workers (backend
+auto-selects "nccl"); these tests pin the choreography: one-shot shared
+staging + broadcast, all-reduce fan-in, SPMD exec, and the
+communicator-rebuild-on-death policy.
+"""
+
+import time
+
+import pytest
+import torch
+
+from fiber_amd.pool import ZPool, ResilientZPool, current_worker_group
+
+
+def _weighted(x, theta=None):
+    # theta arrives via the pool broadcast, not via the chunk payload
+    return float(theta.sum()) * x
+
+
+def _onehot4(x):
+    return torch.full((4,), float(x))
+
+
+def _rank_world():
+    g = current_worker_group()
+    return (g.rank, g.size)
+
+
+def _group_allreduce_rank():
+    g = current_worker_group()
+    t = torch.tensor([float(g.rank + 1)])
+    g.allreduce(t)
+    return float(t[0])
+
+
+def _slow_echo(x):
+    time.sleep(0.05)
+    return x
+
+
+@pytest.fixture
+def cpool():
+    p = ZPool(processes=2, collective=True)
+    yield p
+    p.terminate()
+    p.join()
+
+
+class TestSharedBroadcast:
+    def test_shared_tensor_reaches_every_worker(self, cpool):
+        theta = torch.arange(16, dtype=torch.float32)
+        out = cpool.map(_weighted, range(8), chunksize=1,
+                        shared={"theta": theta})
+        want = [float(theta.sum()) * x for x in range(8)]
+        assert out == want
+
+    def test_shared_requires_collective_pool(self):
+        with ZPool(processes=2) as p:
+            with pytest.raises(ValueError):
+                p.map(_weighted, range(4), shared={"theta": torch.ones(2)})
+
+    def test_resilient_pool_rejects_collective(self):
+        with pytest.raises(ValueError):
+            ResilientZPool(processes=2, collective=True)
+
+
+class TestReduceSum:
+    def test_reduce_matches_plain_map(self, cpool):
+        xs = list(range(10))
+        reduced = cpool.map(_onehot4, xs, chunksize=2, reduce="sum",
+                            reduce_spec=((4,), torch.float32))
+        plain = cpool.map(_onehot4, xs, chunksize=2)
+        want = torch.stack(plain).sum(dim=0)
+        assert torch.equal(reduced, want)
+        assert torch.equal(reduced, torch.full((4,), float(sum(xs))))
+
+    def test_reduce_needs_spec(self, cpool):
+        with pytest.raises(ValueError):
+            cpool.map(_onehot4, range(4), reduce="sum")
+
+
+class TestRunOnAll:
+    def test_one_task_per_rank_ordered(self, cpool):
+        out = cpool.run_on_all(_rank_world)
+        assert out == [(0, 2), (1, 2)]
+
+    def test_worker_group_collectives(self, cpool):
+        # (rank0+1) + (rank1+1) = 3 on every rank
+        out = cpool.run_on_all(_group_allreduce_rank)
+        assert out == [3.0, 3.0]
+
+
+class TestRebuildOnDeath:
+    def test_collective_map_fails_fast_then_recovers(self, monkeypatch):
+        # bound the rendezvous timeout so the rotation converges quickly
+        # (workers inherit the master's env at spawn)
+        monkeypatch.setenv("FAM_PG_TIMEOUT", "5")
+        pool = ZPool(processes=2, collective=True)
+        try:
+            theta = torch.ones(4)
+            res = pool.map_async(_slow_echo, range(60), chunksize=1,
+                                 shared={"theta": theta})
+            # let the map get going, then murder one worker
+            time.sleep(0.6)
+            with pool._worker_lock:
+                victim = next(iter(pool._workers.values()))
+            victim.kill()
+            with pytest.raises(RuntimeError, match="collective"):
+                res.get(60)
+            # the group rebuilt (new generation): SPMD + collectives
+            # work again.  Convergence needs a few attempts: each retry
+            # that overlaps a rank still timing out on the previous
+            # generation fails and rotates once more (bounded by the PG
+            # timeout per attempt).
+            deadline = time.monotonic() + 90
+            while True:
+                try:
+                    out = pool.run_on_all(_group_allreduce_rank,
+                                          timeout=60)
+                    break
+                except Exception:
+                    if time.monotonic() > deadline:
+                        raise
+                    time.sleep(0.5)
+            assert out == [3.0, 3.0]
+        finally:
+            pool.terminate()
+            pool.join()
