@@ -33,6 +33,14 @@ def main():
     parser.add_argument("--model", choices=["mlp", "conv"], default="mlp",
                         help="flagship MLP config (default) or the "
                              "ConvNet pixel-policy config")
+    parser.add_argument("--mode", choices=["engine", "pool"],
+                        default="engine",
+                        help="engine: one rank per GPU drives ESEngine "
+                             "directly (the driver's torchrun contract); "
+                             "pool: the SAME workload dispatched through "
+                             "the framework — Pool(collective=True) "
+                             "SPMD fan-out + pool-communicator "
+                             "collectives (single-process launch only)")
     args = parser.parse_args()
     if args.pop_per_gpu is None:
         # Big shards are the MI355X-first choice (288 GB HBM, fixed costs
@@ -43,6 +51,9 @@ def main():
         args.horizon = 256 if args.model == "mlp" else 64
 
     import torch
+
+    if args.mode == "pool":
+        return bench_through_pool(args)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -131,6 +142,65 @@ def main():
 
     if ctx is not None:
         ctx.shutdown()
+
+
+def bench_through_pool(args):
+    """The same flagship metric, dispatched THROUGH the framework:
+    Pool(collective=True) with one GPU-pinned worker per device, SPMD
+    fan-out per block of iterations, fitness/grad collectives on the
+    pool communicator over xGMI.  Overhead vs engine mode is the cost of
+    the framework (measured: profiles/r02_es_pool_overhead.md)."""
+    import json as _json
+    import time as _time
+
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        raise SystemExit("--mode pool is a single-process launch: the "
+                         "pool spawns its own per-GPU workers")
+    from fiber_amd.pool import ZPool
+    from examples.es_pool import es_steps, init_es_worker
+
+    cfg = dict(pop_per_gpu=args.pop_per_gpu, horizon=args.horizon)
+    pool = ZPool(processes=args.gpus, gpu_per_worker=1, collective=True,
+                 initializer=init_es_worker, initargs=(cfg, args.model))
+    try:
+        pool.run_on_all(es_steps, (0, args.warmup, True), timeout=1800)
+        t0 = _time.perf_counter()
+        per_rank = pool.run_on_all(
+            es_steps, (args.warmup, args.steps, True), timeout=3600
+        )
+        elapsed = _time.perf_counter() - t0
+    finally:
+        pool.terminate()
+        pool.join()
+
+    envs = 64 if args.model == "mlp" else 16
+    rollouts_per_step = args.pop_per_gpu * args.gpus * envs
+    value = rollouts_per_step * args.steps / elapsed
+    print(_json.dumps({
+        "metric": "es_rollouts_per_sec",
+        "value": value,
+        "unit": "rollouts/s",
+        "n_gpus": args.gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if args.model == "mlp" else "bf16+fp8(e4m3)",
+        "data": "synthetic",
+        "config": {
+            "model": ("es-mlp-obs4-h64x64-act2" if args.model == "mlp"
+                      else "es-conv-84x84x4-dqn-act6"),
+            "global_batch": args.pop_per_gpu * args.gpus * envs,
+            "seq_len": args.horizon,
+            "parallelism": "pool-dp%d" % args.gpus,
+            "pop_per_gpu": args.pop_per_gpu,
+            "envs_per_member": envs,
+            "dispatch": "Pool.run_on_all + pool RCCL communicator",
+            "theta_sum": per_rank[0]["theta_sum"],
+        },
+    }))
 
 
 if __name__ == "__main__":
