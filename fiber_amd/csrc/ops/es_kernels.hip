@@ -68,14 +68,23 @@ __device__ inline float fast_tanh(float x) {
 #define OFF_B3 (OFF_W3 + ACT * HID)         // 4608
 #define NPARAMS (OFF_B3 + ACT)              // 4610
 
-#define S1 40  // LDS stride (bf16) for K=32 operands
+// Layer-1 operands only carry 8 k-slots (obs 4 + 4 zeros): the MFMA is
+// K=32, but the kgrp>0 lane groups feed constant-zero fragments instead
+// of reading LDS zeros — identical math, and the w1/xb tiles shrink from
+// 5 KB to 1 KB each.  That drops the block from ~34 KB to ~25 KB of LDS:
+// 6 workgroups/CU instead of 4 on a VALU-latency-bound kernel.
+#define S1 8   // LDS stride (bf16) for layer-1 operands (16 B rows)
 #define S2 72  // LDS stride (bf16) for K=64 operands
 
 // One wave computes a 16-row strip of C[64][64] = A[64][K] x B[K][64],
 // reading A as [row][k] (stride AS) and B as [col][k] (stride BS), then
 // applies bias+tanh and stores C transposed into out[col][row] (stride OS)
 // so the output is ready as the next layer's B-operand.
-template <int K, int AS, int BS, int OS>
+//
+// NARROW=true: the operands carry only 8 real k-slots (obs 4 + 4 zero
+// pad); the K=32 MFMA's kgrp>0 lane groups use constant-zero fragments —
+// bit-identical result, 1/4 the fragment LDS traffic, 1/4 the tile LDS.
+template <int K, int AS, int BS, int OS, bool NARROW = false>
 __device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
                                        const __hip_bfloat16* __restrict__ B,
                                        const float* __restrict__ bias,
@@ -84,12 +93,15 @@ __device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
   const int r0 = wave * 16;
   const int arow = r0 + (lane & 15);
   const int kgrp = lane >> 4;  // 0..3
+  const bf16x8 zfrag = {};
 
   bf16x8 afrag[K / 32];
 #pragma unroll
   for (int kk = 0; kk < K / 32; ++kk) {
-    afrag[kk] = *reinterpret_cast<const bf16x8*>(
-        &A[arow * AS + kk * 32 + kgrp * 8]);
+    afrag[kk] = (NARROW && kgrp != 0)
+                    ? zfrag
+                    : *reinterpret_cast<const bf16x8*>(
+                          &A[arow * AS + (NARROW ? 0 : kk * 32 + kgrp * 8)]);
   }
 
 #pragma unroll
@@ -98,8 +110,11 @@ __device__ inline void mfma_strip_tanh(const __hip_bfloat16* __restrict__ A,
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int kk = 0; kk < K / 32; ++kk) {
-      bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-          &B[bcol * BS + kk * 32 + kgrp * 8]);
+      bf16x8 bfrag =
+          (NARROW && kgrp != 0)
+              ? zfrag
+              : *reinterpret_cast<const bf16x8*>(
+                    &B[bcol * BS + (NARROW ? 0 : kk * 32 + kgrp * 8)]);
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bfrag, acc,
                                                     0, 0, 0);
     }
@@ -245,12 +260,12 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     }
   }
   // zero the K-padding of W1 (A-side zeros make pad products zero) and xb
-  for (int idx = tid; idx < HID * (KPAD - OBS); idx += blockDim.x) {
-    L.pol.w1[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+  for (int idx = tid; idx < HID * (S1 - OBS); idx += blockDim.x) {
+    L.pol.w1[idx / (S1 - OBS)][OBS + idx % (S1 - OBS)] =
         __float2bfloat16(0.f);
   }
-  for (int idx = tid; idx < ENVS * (KPAD - OBS); idx += blockDim.x) {
-    L.xb[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+  for (int idx = tid; idx < ENVS * (S1 - OBS); idx += blockDim.x) {
+    L.xb[idx / (S1 - OBS)][OBS + idx % (S1 - OBS)] =
         __float2bfloat16(0.f);
   }
   // init env state (member-independent: common random numbers), zero accs
@@ -305,8 +320,9 @@ es_rollout_mlp(const float* __restrict__ theta, float sigma, uint32_t seed,
     const int cur = t & 1;
     __syncthreads();
     // ---- phase B: h1 = tanh(W1 x + b1)  (MFMA, K=32) ------------------
-    mfma_strip_tanh<KPAD, S1, S1, S2>(&L.pol.w1[0][0], &L.xb[0][0],
-                                      L.pol.b1, &L.h1[0][0], wave, lane);
+    mfma_strip_tanh<KPAD, S1, S1, S2, true>(&L.pol.w1[0][0], &L.xb[0][0],
+                                            L.pol.b1, &L.h1[0][0], wave,
+                                            lane);
     __syncthreads();
     // ---- phase C: h2 stays in registers; logits fused (MFMA, K=64) ----
     mfma_strip_logits<HID, S2, S2>(&L.pol.w2[0][0], &L.h1[0][0],
@@ -452,17 +468,17 @@ mlp_policy_forward(const float* __restrict__ theta,
   const int row0 = blockIdx.x * ENVS;
 
   for (int j = tid; j < NPARAMS; j += blockDim.x) store_param(&pol, j, theta[j]);
-  for (int idx = tid; idx < HID * (KPAD - OBS); idx += blockDim.x)
-    pol.w1[idx / (KPAD - OBS)][OBS + idx % (KPAD - OBS)] =
+  for (int idx = tid; idx < HID * (S1 - OBS); idx += blockDim.x)
+    pol.w1[idx / (S1 - OBS)][OBS + idx % (S1 - OBS)] =
         __float2bfloat16(0.f);
-  for (int idx = tid; idx < ENVS * KPAD; idx += blockDim.x) {
-    int e = idx / KPAD, d = idx % KPAD;
+  for (int idx = tid; idx < ENVS * S1; idx += blockDim.x) {
+    int e = idx / S1, d = idx % S1;
     float v = (d < OBS && row0 + e < batch) ? X[(row0 + e) * OBS + d] : 0.f;
     xb[e][d] = __float2bfloat16(v);
   }
   __syncthreads();
-  mfma_strip_tanh<KPAD, S1, S1, S2>(&pol.w1[0][0], &xb[0][0], pol.b1,
-                                    &h1[0][0], wave, lane);
+  mfma_strip_tanh<KPAD, S1, S1, S2, true>(&pol.w1[0][0], &xb[0][0], pol.b1,
+                                          &h1[0][0], wave, lane);
   __syncthreads();
   mfma_strip_tanh<HID, S2, S2, S2>(&pol.w2[0][0], &h1[0][0], pol.b2,
                                    &h2[0][0], wave, lane);
