@@ -1,0 +1,155 @@
+"""Minutes-long chaos churn (VERDICT round-1 #7).
+
+Continuous SIGKILL of pool workers / ring producers at random intervals
+catches processes in every transport state: holding the robust ring
+mutex, mid-`send_many` burst (reserved-uncommitted records), mid-recv,
+mid-spawn handshake, and holding a REQ/REP reply ring.  The pool must
+keep returning complete, correct results; the ring must never deliver a
+corrupt payload or wedge.
+
+CI runs a bounded version (~25 s); `FAM_SOAK=1` stretches it to minutes
+(results of a full run: profiles/r02_chaos_soak.md).
+"""
+
+import hashlib
+import os
+import random
+import threading
+import time
+
+from fiber_amd.pool import ResilientZPool
+from fiber_amd.process import Process
+from fiber_amd.transport import ShmRing, new_address
+
+
+def _churn_task(x):
+    time.sleep(0.001 * (x % 5))
+    return x * 3
+
+
+def _soak_seconds(default):
+    return float(os.environ.get("FAM_SOAK_S", "0")) or (
+        180.0 if os.environ.get("FAM_SOAK") else default
+    )
+
+
+class TestPoolChurn:
+    def test_pool_survives_continuous_worker_murder(self):
+        duration = _soak_seconds(20.0)
+        pool = ResilientZPool(processes=4)
+        stop = threading.Event()
+        kills = [0]
+
+        def killer():
+            while not stop.is_set():
+                time.sleep(random.uniform(0.02, 0.12))
+                with pool._worker_lock:
+                    procs = [p for p in pool._workers.values()
+                             if p.exitcode is None]
+                if procs:
+                    try:
+                        random.choice(procs).kill()
+                        kills[0] += 1
+                    except Exception:
+                        pass
+
+        th = threading.Thread(target=killer, daemon=True)
+        th.start()
+        deadline = time.monotonic() + duration
+        rounds = 0
+        try:
+            while time.monotonic() < deadline:
+                n = 300
+                res = pool.map(_churn_task, range(n), chunksize=3)
+                assert res == [x * 3 for x in range(n)], \
+                    "lost/corrupt results in churn round %d" % rounds
+                rounds += 1
+        finally:
+            stop.set()
+            th.join(2)
+            pool.terminate()
+            pool.join()
+        assert rounds >= 2, "soak made no progress"
+        assert kills[0] >= 5, "killer thread barely fired (%d)" % kills[0]
+
+
+def _burst_producer(addr, seed):
+    """Send checksummed payloads in send_many bursts forever (until
+    killed): the kill lands inside reserve/copy/commit with high
+    probability."""
+    rng = random.Random(seed)
+    ring = ShmRing(addr, False, 1 << 20, 20.0)
+    i = 0
+    while True:
+        burst = []
+        for _ in range(rng.randint(1, 24)):
+            body = os.urandom(rng.randint(0, 3000))
+            digest = hashlib.blake2b(body, digest_size=8).digest()
+            burst.append(digest + body)
+            i += 1
+        ring.send_many(burst, timeout=5.0)
+
+
+class TestRingChurn:
+    def test_ring_integrity_under_producer_murder(self):
+        duration = _soak_seconds(12.0)
+        addr = new_address("fam-soak")
+        ring = ShmRing(addr, True, 1 << 20, 20.0)
+        procs = []
+        try:
+            for k in range(3):
+                p = Process(target=_burst_producer, args=(addr, k),
+                            name="soak-prod-%d" % k)
+                p.start()
+                procs.append(p)
+            deadline = time.monotonic() + duration
+            received = 0
+            kills = 0
+            rng = random.Random(7)
+            next_kill = time.monotonic() + rng.uniform(0.05, 0.3)
+            while time.monotonic() < deadline:
+                for payload in ring.recv_many(64, 0.2):
+                    digest, body = payload[:8], payload[8:]
+                    assert hashlib.blake2b(
+                        body, digest_size=8
+                    ).digest() == digest, "corrupt payload after kill"
+                    received += 1
+                if time.monotonic() >= next_kill:
+                    victim = rng.choice(procs)
+                    victim.kill()
+                    victim.join(5)
+                    idx = procs.index(victim)
+                    p = Process(target=_burst_producer,
+                                args=(addr, 100 + kills),
+                                name="soak-prod-r%d" % kills)
+                    p.start()
+                    procs[idx] = p
+                    kills += 1
+                    next_kill = time.monotonic() + rng.uniform(0.05, 0.3)
+            assert received > 1000, "soak consumed too little (%d)" % received
+            assert kills >= 10, "too few producer kills (%d)" % kills
+            # the ring must still be fully usable afterwards
+            for p in procs:
+                p.kill()
+                p.join(5)
+            # drain EVERYTHING left, verifying integrity
+            while True:
+                leftovers = ring.recv_many(128, 0.5)
+                if not leftovers:
+                    break
+                for payload in leftovers:
+                    digest, body = payload[:8], payload[8:]
+                    assert hashlib.blake2b(
+                        body, digest_size=8
+                    ).digest() == digest
+            ring.send(b"post-soak", 5.0)
+            assert ring.recv(5.0) == b"post-soak"
+        finally:
+            for p in procs:
+                try:
+                    p.kill()
+                    p.join(5)
+                except Exception:
+                    pass
+            ring.close()
+            ring.unlink()
