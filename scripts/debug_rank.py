@@ -1,4 +1,6 @@
 """One-off: diagnose sorted centered-rank mismatch at n=131072."""
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from fiber_amd import ops
 
