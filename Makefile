@@ -15,3 +15,8 @@ bench: build
 
 clean:
 	rm -rf build fiber_amd/*.so
+
+tsan:  ## ThreadSanitizer run over the transport engine (no GPU needed)
+	g++ -std=c++17 -O1 -g -fsanitize=thread -DFAM_NO_PYBIND \
+	  fiber_amd/csrc/tsan_harness.cpp -o /tmp/fam_tsan -lpthread -lrt
+	/tmp/fam_tsan

@@ -32,7 +32,11 @@
 // Records: [u32 len][u32 state][u32 writer_pid][u32 pad][payload..] padded
 // to 8 B.  A WRAP marker (len == 0xFFFFFFFF) means "skip to offset 0".
 
+// FAM_NO_PYBIND: built as a plain C++ TU by the TSAN harness
+// (csrc/tsan_harness.cpp) — same engine code, no Python linkage.
+#ifndef FAM_NO_PYBIND
 #include <pybind11/pybind11.h>
+#endif
 
 #include <atomic>
 #include <cerrno>
@@ -54,7 +58,9 @@
 #include <time.h>
 #include <unistd.h>
 
+#ifndef FAM_NO_PYBIND
 namespace py = pybind11;
+#endif
 
 namespace {
 
@@ -887,6 +893,7 @@ class ShmRing {
 
 }  // namespace
 
+#ifndef FAM_NO_PYBIND
 PYBIND11_MODULE(_transport, m) {
   m.doc() = "fiber_amd shared-memory message transport (single MI355X node)";
 
@@ -992,3 +999,4 @@ PYBIND11_MODULE(_transport, m) {
       .def_property_readonly("is_owner", &ShmRing::is_owner)
       .def_property_readonly("name", &ShmRing::name);
 }
+#endif  // FAM_NO_PYBIND

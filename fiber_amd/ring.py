@@ -32,7 +32,8 @@ class RingNode:
         self.rank = rank
         self.ip = "127.0.0.1"
         self.port = None
-        self.connected = False
+        self.connected = False  # set from REAL child feedback (run())
+        self.pid = None
 
 
 class RingContext:
@@ -206,18 +207,36 @@ class RingContext:
 
 
 def _ring_target(rank, size, func_blob, init_blob, master_port, backend,
-                 bucket_mb):
+                 bucket_mb, status_addr=None):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ["MASTER_PORT"] = str(master_port)
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(size)
     ctx = RingContext(rank, size, backend=backend, bucket_mb=bucket_mb)
+    status = None
+    if status_addr is not None:
+        from .queues import SimpleQueue
+
+        status = SimpleQueue(status_addr, _create=False)
+        # real membership feedback (the reference kept a Manager-shared
+        # rendezvous table, fiber/experimental/ring.py:44-98): "up" once
+        # the rank is actually executing, before any collective
+        status.put(("up", rank, os.getpid()))
     initializer = serialization.loads(init_blob) if init_blob else None
     func = serialization.loads(func_blob)
     try:
         if initializer is not None:
             initializer(ctx)
         func(rank, size)
+        if status is not None:
+            status.put(("done", rank, 0))
+    except BaseException:
+        if status is not None:
+            try:
+                status.put(("failed", rank, 1))
+            except Exception:
+                pass
+        raise
     finally:
         ctx.shutdown()
 
@@ -241,6 +260,8 @@ class Ring:
         self._procs = []
 
     def run(self, timeout=None):
+        from .queues import SimpleQueue
+
         master_port = _free_tcp_port()
         func_blob = serialization.dumps_closure(self.func)
         init_blob = (
@@ -248,6 +269,7 @@ class Ring:
             if self.initializer
             else None
         )
+        status = SimpleQueue()
 
         meta = {"gpu": self.gpu_per_rank} if self.gpu_per_rank else {}
         for rank in range(self.size):
@@ -260,6 +282,7 @@ class Ring:
                     master_port=master_port,
                     backend=self.backend,
                     bucket_mb=self.bucket_mb,
+                    status_addr=status._addr,
                 ),
                 meta,
             )
@@ -267,13 +290,48 @@ class Ring:
             proc.start()
             self._procs.append(proc)
             self.members[rank].port = master_port
-            self.members[rank].connected = True
+
+        # Membership from REAL child feedback: a rank that never reports
+        # "up" within the rendezvous window is named in the error instead
+        # of surfacing as an opaque join timeout.
+        import time as _time
+
+        up_deadline = _time.monotonic() + float(
+            os.environ.get("FAM_PG_TIMEOUT", "600")
+        )
+        pending_up = set(range(self.size))
+        while pending_up:
+            if any(p.exitcode not in (0, None) for p in self._procs):
+                break  # a rank already died; fall through to join/report
+            if _time.monotonic() > up_deadline:
+                for p in self._procs:
+                    p.terminate()
+                raise RuntimeError(
+                    "ring ranks never came up: %s" % sorted(pending_up)
+                )
+            try:
+                kind, rank, info = status.get(timeout=0.2)
+            except TimeoutError:
+                continue
+            if kind == "up":
+                self.members[rank].connected = True
+                self.members[rank].pid = info
+                pending_up.discard(rank)
 
         failures = []
         for proc in self._procs:
             proc.join(timeout)
             if proc.exitcode not in (0, None):
                 failures.append((proc.name, proc.exitcode))
+        # drain terminal status records (best effort observability)
+        try:
+            while True:
+                kind, rank, _info = status.get_nowait()
+                if kind in ("done", "failed"):
+                    self.members[rank].connected = False
+        except Exception:
+            pass
+        status.close()
         if failures:
             for proc in self._procs:
                 proc.terminate()

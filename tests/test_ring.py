@@ -10,6 +10,10 @@ from fiber_amd.queues import SimpleQueue
 from fiber_amd.ring import Ring, RingContext, RingNode
 
 
+def _noop_rank(rank, size):
+    return None
+
+
 def _allreduce_work(rank, size, out=None):
     ctx = RingContext(rank, size, backend="gloo")
     t = torch.ones(100) * (rank + 1)
@@ -88,6 +92,17 @@ class TestRing:
         ring = Ring(3, lambda r, s: None, gpu_per_rank=0)
         assert [m.rank for m in ring.members] == [0, 1, 2]
         assert isinstance(ring.members[0], RingNode)
+
+    def test_membership_from_real_child_feedback(self):
+        """Every member's pid comes from the rank's own 'up' report —
+        not optimistic bookkeeping.  After a clean
+        run the nodes report done and are marked disconnected."""
+        ring = Ring(2, _noop_rank, backend="gloo", gpu_per_rank=0)
+        ring.run(timeout=300)
+        pids = [m.pid for m in ring.members]
+        assert all(isinstance(p, int) and p > 0 for p in pids)
+        assert len(set(pids)) == 2
+        assert all(m.connected is False for m in ring.members)
 
     def test_failed_rank_raises(self):
         def boom(rank, size):
