@@ -174,8 +174,11 @@ static void launch_perturb(uintptr_t theta, int nparams, int np_pad,
                            int member_offset, int pop, uintptr_t wpert,
                            uintptr_t w3_fp8, uintptr_t w1_fp8,
                            uintptr_t stream) {
-  const int bx = 64;  // grid-stride over param blocks
-  hipLaunchKernelGGL(es_perturb, dim3(bx, pop), dim3(256), 0,
+  if ((member_offset | pop) & 1)
+    throw std::runtime_error(
+        "es_perturb needs pair-aligned member_offset/pop");
+  const int bx = 64;  // grid-stride over param blocks; grid.y = PAIRS
+  hipLaunchKernelGGL(es_perturb, dim3(bx, pop / 2), dim3(256), 0,
                      (hipStream_t)stream, (const float*)theta, nparams,
                      np_pad, (float)sigma, seed, (const uint32_t*)iterp,
                      member_offset, (__hip_bfloat16*)wpert,

@@ -78,15 +78,22 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
                                       __hip_bfloat16* __restrict__ wpert,
                                       unsigned char* __restrict__ w3_fp8,
                                       unsigned char* __restrict__ w1_fp8) {
+  // One block materializes BOTH members of an antithetic pair: the
+  // Philox + Box-Muller draw (the kernel's measured bottleneck,
+  // NOTES_ROUND2 #2) runs once per pair instead of once per member.
+  // Identical outputs: member 2k writes theta + sigma*eps, member 2k+1
+  // writes theta - sigma*eps from the same eps.
   const uint32_t iter = *iterp;
-  const int member = member_offset + blockIdx.y;
-  const uint32_t pair = (uint32_t)(member >> 1);
-  const float sgn = (member & 1) ? -sigma : sigma;
-  __hip_bfloat16* out = wpert + (size_t)blockIdx.y * np_pad;
-  unsigned char* out8 =
-      w3_fp8 ? w3_fp8 + (size_t)blockIdx.y * (FCU * NFLAT) : nullptr;
-  unsigned char* out8w1 =
-      w1_fp8 ? w1_fp8 + (size_t)blockIdx.y * (C1 * 256) : nullptr;
+  const int member0 = member_offset + 2 * blockIdx.y;  // even member
+  const uint32_t pair = (uint32_t)(member0 >> 1);
+  __hip_bfloat16* outp = wpert + (size_t)(2 * blockIdx.y) * np_pad;
+  __hip_bfloat16* outm = outp + np_pad;
+  unsigned char* out8p =
+      w3_fp8 ? w3_fp8 + (size_t)(2 * blockIdx.y) * (FCU * NFLAT) : nullptr;
+  unsigned char* out8m = out8p ? out8p + (FCU * NFLAT) : nullptr;
+  unsigned char* out8w1p =
+      w1_fp8 ? w1_fp8 + (size_t)(2 * blockIdx.y) * (C1 * 256) : nullptr;
+  unsigned char* out8w1m = out8w1p ? out8w1p + (C1 * 256) : nullptr;
   const int jb0 = blockIdx.x * blockDim.x + threadIdx.x;
   const int stride = gridDim.x * blockDim.x;
   for (int jb = jb0; jb * 4 < nparams; jb += stride) {
@@ -97,15 +104,22 @@ extern "C" __global__ void es_perturb(const float* __restrict__ theta,
     for (int u = 0; u < 4; ++u) {
       const int j = j0 + u;
       if (j < nparams) {
-        const float v = theta[j] + sgn * z[u];
-        out[j] = __float2bfloat16(v);
+        const float sz = sigma * z[u];
+        const float vp = theta[j] + sz;
+        const float vm = theta[j] - sz;
+        outp[j] = __float2bfloat16(vp);
+        outm[j] = __float2bfloat16(vm);
         // the fc layer consumes W3 in OCP e4m3 (fp8) — halves the
         // dominant HBM term of the rollout (profiles: conv_fc is
         // BW-bound on single-use weights)
-        if (out8 && j >= COFF_W3 && j < COFF_B3)
-          out8[j - COFF_W3] = __hip_fp8_e4m3(v).__x;
-        if (out8w1 && j < COFF_B1)
-          out8w1[j] = __hip_fp8_e4m3(v).__x;
+        if (out8p && j >= COFF_W3 && j < COFF_B3) {
+          out8p[j - COFF_W3] = __hip_fp8_e4m3(vp).__x;
+          out8m[j - COFF_W3] = __hip_fp8_e4m3(vm).__x;
+        }
+        if (out8w1p && j < COFF_B1) {
+          out8w1p[j] = __hip_fp8_e4m3(vp).__x;
+          out8w1m[j] = __hip_fp8_e4m3(vm).__x;
+        }
       }
     }
   }

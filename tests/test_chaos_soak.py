@@ -132,18 +132,23 @@ class TestRingChurn:
             for p in procs:
                 p.kill()
                 p.join(5)
-            # drain EVERYTHING left, verifying integrity
+            # Liveness + integrity after the massacre: enqueue a
+            # checksummed sentinel and drain until it arrives.  (A plain
+            # drain-then-roundtrip races with dead-writer reclaim: a
+            # reserved record from a just-killed producer can resolve
+            # AFTER an empty recv_many, re-exposing leftovers.)
+            body = b"post-soak"
+            sentinel = hashlib.blake2b(body, digest_size=8).digest() + body
+            ring.send(sentinel, 5.0)
             while True:
-                leftovers = ring.recv_many(128, 0.5)
-                if not leftovers:
+                payload = ring.recv(10.0)
+                assert payload is not None, "ring wedged post-soak"
+                digest, pbody = payload[:8], payload[8:]
+                assert hashlib.blake2b(
+                    pbody, digest_size=8
+                ).digest() == digest, "corrupt leftover after massacre"
+                if pbody == body:
                     break
-                for payload in leftovers:
-                    digest, body = payload[:8], payload[8:]
-                    assert hashlib.blake2b(
-                        body, digest_size=8
-                    ).digest() == digest
-            ring.send(b"post-soak", 5.0)
-            assert ring.recv(5.0) == b"post-soak"
         finally:
             for p in procs:
                 try:
