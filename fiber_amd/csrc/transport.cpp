@@ -64,7 +64,13 @@ namespace py = pybind11;
 
 namespace {
 
-constexpr uint32_t kMagic = 0xFA3B71A8u;  // bumped: spill-flag header
+constexpr uint32_t kMagic = 0xFA3B71A9u;  // bumped: spill accounting
+// Outstanding spill segments per ring are capped: spilled payloads
+// bypass the ring-capacity backpressure, so without this a fast
+// producer of huge messages could fill /dev/shm (RAM) unboundedly.
+// 16 in flight preserves pipelining; producers block (normal not-full
+// wait) once the cap is reached.
+constexpr uint64_t kMaxSpillSegs = 16;
 constexpr uint32_t kWrapMarker = 0xFFFFFFFFu;
 constexpr size_t kAlign = 8;
 
@@ -93,6 +99,7 @@ struct Header {
   uint64_t msg_count;  // messages currently in the ring
   uint64_t total_in;   // lifetime enqueued messages
   uint64_t total_out;  // lifetime dequeued messages
+  uint64_t spill_count;  // outstanding spill segments (backpressure)
   uint32_t closed;
 };
 
@@ -421,7 +428,9 @@ class ShmRing {
           hdr_->head = hdr_->tail = 0;
         uint64_t tail = hdr_->tail;
         uint64_t waste = (tail + need > cap) ? (cap - tail) : 0;
-        if (cap - hdr_->used >= need + waste) {
+        const bool spill_ok = !(flags & kFlagSpill) ||
+                              hdr_->spill_count < kMaxSpillSegs;
+        if (spill_ok && cap - hdr_->used >= need + waste) {
           if (waste) {
             uint32_t marker = kWrapMarker;
             std::memcpy(data_ + tail, &marker, 4);
@@ -440,6 +449,7 @@ class ShmRing {
           hdr_->used += need;
           hdr_->msg_count += 1;
           hdr_->total_in += 1;
+          if (flags & kFlagSpill) hdr_->spill_count += 1;
           break;
         }
         if (timeout == 0) return false;
@@ -526,6 +536,9 @@ class ShmRing {
             // exact wrap-waste pricing — see send()
             uint64_t waste = (tail + need > cap) ? (cap - tail) : 0;
             if (cap - hdr_->used < need + waste) break;
+            if ((flags[done + burst] & kFlagSpill) &&
+                hdr_->spill_count >= kMaxSpillSegs)
+              break;  // spill budget backpressure
             if (waste) {
               uint32_t marker = kWrapMarker;
               std::memcpy(data_ + tail, &marker, 4);
@@ -544,6 +557,7 @@ class ShmRing {
             hdr_->used += need;
             hdr_->msg_count += 1;
             hdr_->total_in += 1;
+            if (flags[done + burst] & kFlagSpill) hdr_->spill_count += 1;
             ++burst;
           }
           if (burst > 0) break;
@@ -611,9 +625,11 @@ class ShmRing {
       spill_read(nm, real_len ? &(*out)[0] : nullptr, real_len);
     } catch (...) {
       consume_record(head, len32);
+      if (hdr_->spill_count) hdr_->spill_count -= 1;
       throw;
     }
     consume_record(head, len32);
+    if (hdr_->spill_count) hdr_->spill_count -= 1;
   }
 
   void consume_record(uint64_t head, uint32_t len32) {
@@ -624,11 +640,14 @@ class ShmRing {
     hdr_->total_out += 1;
   }
 
-  void reclaim_dead_record(uint64_t head, uint32_t len32) {
+  void reclaim_dead_record(uint64_t head, uint32_t len32, uint32_t flags) {
     size_t need = record_bytes(len32);
     hdr_->head = (head + need) % hdr_->capacity;
     hdr_->used -= need;
     hdr_->msg_count -= 1;
+    // flags were written under the reserve lock, so they are valid even
+    // though the payload is not: release the dead writer's spill budget
+    if ((flags & kFlagSpill) && hdr_->spill_count) hdr_->spill_count -= 1;
   }
 
   // Shared blocking structure for recv / recv_into / peek.
@@ -665,7 +684,7 @@ class ShmRing {
             uint32_t pid;
             std::memcpy(&pid, data_ + head + 8, 4);
             if (!process_alive(pid)) {
-              reclaim_dead_record(head, len32);
+              reclaim_dead_record(head, len32, flags);
               lock.unlock();
               bump_and_wake(&hdr_->fut_not_full,
                             &hdr_->waiters_not_full);
@@ -743,7 +762,7 @@ class ShmRing {
             uint32_t pid;
             std::memcpy(&pid, data_ + head + 8, 4);
             if (!process_alive(pid)) {
-              reclaim_dead_record(head, len32);
+              reclaim_dead_record(head, len32, flags);
               lock.unlock();
               bump_and_wake(&hdr_->fut_not_full,
                             &hdr_->waiters_not_full);
@@ -808,9 +827,12 @@ class ShmRing {
               spill_read(nm, buf, real_len);
             } catch (...) {
               consume_record(head, len32);
+              if (hdr_->spill_count) hdr_->spill_count -= 1;
               throw;
             }
             consume_record(head, len32);
+            // release the spill budget (same as consume_spill)
+            if (hdr_->spill_count) hdr_->spill_count -= 1;
             result = (int64_t)real_len;
             return true;
           }

@@ -41,7 +41,7 @@ uint64_t fnv1a(const char* p, size_t n) {
 int main() {
   const char* name = "fam-tsan-ring";
   const int kProducers = 4, kConsumers = 4;
-  const int kPerProducer = 4000;
+  const int kPerProducer = 1500;
   ShmRing ring(name, true, 64 << 10, 5.0);
 
   std::atomic<long> consumed{0}, produced{0};
@@ -70,12 +70,15 @@ int main() {
           make_msg(seed + 1, &b);
           const char* bufs[2] = {a.data(), b.data()};
           size_t lens[2] = {a.size(), b.size()};
-          size_t sent = ring.send_many(bufs, lens, 2, 30.0);
+          // blocking sends: under TSAN's slowdown + spill backpressure a
+          // finite timeout could drop messages and wedge the consumers'
+          // consumed==total exit condition
+          size_t sent = ring.send_many(bufs, lens, 2, -1.0);
           produced += (long)sent;
           ++i;  // consumed two seeds
         } else {
           make_msg(seed, &msg);
-          if (ring.send(msg.data(), msg.size(), 30.0)) ++produced;
+          if (ring.send(msg.data(), msg.size(), -1.0)) ++produced;
         }
       }
     });

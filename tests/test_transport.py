@@ -103,6 +103,28 @@ class TestShmRing:
             ring.close()
             ring.unlink()
 
+    def test_spill_budget_backpressure(self):
+        """Spilled payloads bypass ring-capacity backpressure, so the
+        number of outstanding spill segments is capped (16): a producer
+        of huge messages blocks instead of filling /dev/shm."""
+        name = new_address("fam-t")
+        ring = ShmRing(name, True, 8 << 10, 5.0)
+        try:
+            big = os.urandom(64 << 10)
+            sent = 0
+            while ring.send(big, 0.0):
+                sent += 1
+            assert sent == 16, sent
+            assert len(_spill_segments(name)) == 16
+            assert ring.recv(1.0) == big  # frees one budget slot
+            assert ring.send(big, 1.0)
+            for _ in range(16):
+                assert ring.recv(1.0) == big
+            assert not _spill_segments(name)
+        finally:
+            ring.close()
+            ring.unlink()
+
     def test_spill_sweep_on_unlink(self):
         """Unread spill segments are swept when the ring owner unlinks."""
         name = new_address("fam-t")
