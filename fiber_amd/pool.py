@@ -1171,6 +1171,11 @@ class ZPool:
                     "worker that drew no chunk must still contribute "
                     "zeros of the right shape to the all-reduce"
                 )
+            if reduce is not None and (callback or error_callback):
+                raise ValueError(
+                    "callbacks are not supported with reduce= (use "
+                    "ReduceResult.get/wait)"
+                )
             flags |= _FLAG_COLLECTIVE
         items = list(iterable)
         n = len(items)
@@ -1186,6 +1191,12 @@ class ZPool:
             self._begin_collective(seq, shared, reduce, reduce_spec)
         if n == 0:
             self._inventory.put(seq, 0, [], None)
+            if reduce is not None:
+                # no chunks will arrive, so fire the fan-in directly:
+                # every worker contributes zeros and rank 0 returns the
+                # zero tensor
+                self._reduce_pending.pop(seq, None)
+                self._ctl_broadcast(lambda rank, ident: ("reduce_go", seq))
             if callback is not None:
                 result._fire_callbacks(None, [])
             return result

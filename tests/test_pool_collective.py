@@ -128,3 +128,16 @@ class TestRebuildOnDeath:
         finally:
             pool.terminate()
             pool.join()
+
+
+class TestReduceEdges:
+    def test_empty_iterable_reduce(self, cpool):
+        out = cpool.map(_onehot4, [], reduce="sum",
+                        reduce_spec=((4,), torch.float32))
+        assert torch.equal(out, torch.zeros(4))
+
+    def test_reduce_rejects_callbacks(self, cpool):
+        with pytest.raises(ValueError, match="callbacks"):
+            cpool.map_async(_onehot4, range(4), reduce="sum",
+                            reduce_spec=((4,), torch.float32),
+                            callback=lambda r: None)
