@@ -8,9 +8,13 @@
 //
 // CRASH SAFETY (the whole point of the design):
 //   * state mutations are guarded by a process-shared ROBUST pthread
-//     mutex — a process dying inside the critical section hands the next
-//     locker EOWNERDEAD + a consistent-enough state (head/tail/used are
-//     mutated only after payload/header writes);
+//     mutex; every multi-field mutation is a roll-forward TRANSACTION
+//     (Header::txn): post-state journaled + armed before the apply, so
+//     the next EOWNERDEAD locker lands the dead owner's mutation
+//     exactly — SIGKILL between any two field writes cannot skew
+//     used/msg_count/spill accounting (the long chaos soak caught the
+//     earlier "ordering makes it consistent-enough" version leaking
+//     `used` bytes per in-lock kill until the ring wedged);
 //   * blocking uses RAW FUTEX sequence words, NOT pthread condvars:
 //     glibc condvars contain an internal NON-robust lock, so a process
 //     SIGKILLed inside cond_wait's bookkeeping permanently wedges every
@@ -64,7 +68,7 @@ namespace py = pybind11;
 
 namespace {
 
-constexpr uint32_t kMagic = 0xFA3B71A9u;  // bumped: spill accounting
+constexpr uint32_t kMagic = 0xFA3B71AAu;  // bumped: txn journal
 // Outstanding spill segments per ring are capped: spilled payloads
 // bypass the ring-capacity backpressure, so without this a fast
 // producer of huge messages could fill /dev/shm (RAM) unboundedly.
@@ -101,6 +105,16 @@ struct Header {
   uint64_t total_out;  // lifetime dequeued messages
   uint64_t spill_count;  // outstanding spill segments (backpressure)
   uint32_t closed;
+  // Crash-atomicity journal.  Every multi-field state mutation first
+  // writes its POST-state here and arms txn_stage (release), applies,
+  // then disarms (release).  A process SIGKILLed mid-apply leaves
+  // stage==1; the next EOWNERDEAD locker rolls the snapshot forward.
+  // (The long chaos soak proved "mutation order makes partial death
+  // consistent-enough" wrong: a kill between used+= and msg_count+=
+  // leaks `used` bytes forever — ~600 kills wedged an 1 MB ring.)
+  uint32_t txn_stage;
+  uint64_t txn[7];  // head, tail, used, msg_count, total_in, total_out,
+                    // spill_count (post-state)
 };
 
 inline size_t record_bytes(size_t len) {
@@ -261,14 +275,49 @@ inline void spill_sweep(const std::string& ring_name) {
   for (auto& v : victims) shm_unlink(v.c_str());
 }
 
+// Journal helpers: arm writes the intended post-state, apply copies it
+// into the live fields, disarm closes the transaction.  Stage
+// transitions are release stores so a crash at ANY instruction leaves
+// either stage==0 (nothing or everything applied) or stage==1 with a
+// complete snapshot (roll forward).
+inline void txn_arm(Header* h, uint64_t head, uint64_t tail, uint64_t used,
+                    uint64_t msg, uint64_t tin, uint64_t tout,
+                    uint64_t spill) {
+  h->txn[0] = head;
+  h->txn[1] = tail;
+  h->txn[2] = used;
+  h->txn[3] = msg;
+  h->txn[4] = tin;
+  h->txn[5] = tout;
+  h->txn[6] = spill;
+  as_atomic(&h->txn_stage)->store(1, std::memory_order_release);
+}
+
+inline void txn_apply(Header* h) {
+  h->head = h->txn[0];
+  h->tail = h->txn[1];
+  h->used = h->txn[2];
+  h->msg_count = h->txn[3];
+  h->total_in = h->txn[4];
+  h->total_out = h->txn[5];
+  h->spill_count = h->txn[6];
+  as_atomic(&h->txn_stage)->store(0, std::memory_order_release);
+}
+
+inline void txn_recover(Header* h) {
+  if (as_atomic(&h->txn_stage)->load(std::memory_order_acquire) == 1)
+    txn_apply(h);  // roll the dead owner's mutation forward
+}
+
 class RobustLock {
  public:
-  explicit RobustLock(pthread_mutex_t* mu) : mu_(mu) {
+  explicit RobustLock(pthread_mutex_t* mu, Header* hdr = nullptr)
+      : mu_(mu) {
     int rc = pthread_mutex_lock(mu_);
     if (rc == EOWNERDEAD) {
-      // Previous owner died mid-critical-section.  Ring mutations are
-      // ordered so that head/tail/used advance only after payload
-      // writes, so the state is consistent enough to continue.
+      // Previous owner died mid-critical-section: roll its armed
+      // transaction forward (see Header::txn), then mark consistent.
+      if (hdr) txn_recover(hdr);
       pthread_mutex_consistent(mu_);
     } else if (rc != 0) {
       throw std::runtime_error("mutex lock failed: " + std::to_string(rc));
@@ -413,7 +462,7 @@ class ShmRing {
       uint32_t snap = as_atomic(&hdr_->fut_not_full)
                           ->load(std::memory_order_acquire);
       {
-        RobustLock lock(&hdr_->mu);
+        RobustLock lock(&hdr_->mu, hdr_);
         if (hdr_->closed) throw std::runtime_error("ring closed");
         // Exact fit check: a record that would straddle the end costs an
         // extra `cap - tail` wasted bytes (wrap marker + dead space).
@@ -431,10 +480,11 @@ class ShmRing {
         const bool spill_ok = !(flags & kFlagSpill) ||
                               hdr_->spill_count < kMaxSpillSegs;
         if (spill_ok && cap - hdr_->used >= need + waste) {
+          // DATA writes first (not journaled — invisible until the
+          // state transaction lands), then one atomic state txn.
           if (waste) {
             uint32_t marker = kWrapMarker;
             std::memcpy(data_ + tail, &marker, 4);
-            hdr_->used += waste;
             tail = 0;
           }
           rec = tail;
@@ -445,11 +495,12 @@ class ShmRing {
               ->store(kStReserved, std::memory_order_relaxed);
           std::memcpy(data_ + rec + 8, &pid, 4);
           std::memcpy(data_ + rec + 12, &flags, 4);
-          hdr_->tail = (tail + need) % cap;
-          hdr_->used += need;
-          hdr_->msg_count += 1;
-          hdr_->total_in += 1;
-          if (flags & kFlagSpill) hdr_->spill_count += 1;
+          txn_arm(hdr_, hdr_->head, (tail + need) % cap,
+                  hdr_->used + waste + need, hdr_->msg_count + 1,
+                  hdr_->total_in + 1, hdr_->total_out,
+                  hdr_->spill_count +
+                      ((flags & kFlagSpill) ? 1 : 0));
+          txn_apply(hdr_);
           break;
         }
         if (timeout == 0) return false;
@@ -522,7 +573,7 @@ class ShmRing {
         uint32_t snap = as_atomic(&hdr_->fut_not_full)
                             ->load(std::memory_order_acquire);
         {
-          RobustLock lock(&hdr_->mu);
+          RobustLock lock(&hdr_->mu, hdr_);
           if (hdr_->closed) throw std::runtime_error("ring closed");
           uint64_t cap = hdr_->capacity;
           while (done + burst < n && burst < kMaxBatch) {
@@ -542,7 +593,6 @@ class ShmRing {
             if (waste) {
               uint32_t marker = kWrapMarker;
               std::memcpy(data_ + tail, &marker, 4);
-              hdr_->used += waste;
               tail = 0;
             }
             uint32_t len32 = (uint32_t)len;
@@ -553,11 +603,12 @@ class ShmRing {
             std::memcpy(data_ + tail + 8, &pid, 4);
             std::memcpy(data_ + tail + 12, &flags[done + burst], 4);
             recs[burst] = tail;
-            hdr_->tail = (tail + need) % cap;
-            hdr_->used += need;
-            hdr_->msg_count += 1;
-            hdr_->total_in += 1;
-            if (flags[done + burst] & kFlagSpill) hdr_->spill_count += 1;
+            txn_arm(hdr_, hdr_->head, (tail + need) % cap,
+                    hdr_->used + waste + need, hdr_->msg_count + 1,
+                    hdr_->total_in + 1, hdr_->total_out,
+                    hdr_->spill_count +
+                        ((flags[done + burst] & kFlagSpill) ? 1 : 0));
+            txn_apply(hdr_);
             ++burst;
           }
           if (burst > 0) break;
@@ -598,8 +649,11 @@ class ShmRing {
     uint32_t len32;
     std::memcpy(&len32, data_ + head, 4);
     if (len32 == kWrapMarker) {
-      hdr_->used -= cap - head;
-      hdr_->head = head = 0;
+      txn_arm(hdr_, 0, hdr_->tail, hdr_->used - (cap - head),
+              hdr_->msg_count, hdr_->total_in, hdr_->total_out,
+              hdr_->spill_count);
+      txn_apply(hdr_);
+      head = 0;
       std::memcpy(&len32, data_ + head, 4);
     }
     uint32_t state =
@@ -617,37 +671,39 @@ class ShmRing {
     uint64_t real_len;
     std::string nm;
     if (!spill_ctrl_decode(data_ + head + kRecHdr, len32, &real_len, &nm)) {
-      consume_record(head, len32);
+      consume_record(head, len32, kFlagSpill);
       throw std::runtime_error("corrupt spill control record");
     }
     out->resize(real_len);
     try {
       spill_read(nm, real_len ? &(*out)[0] : nullptr, real_len);
     } catch (...) {
-      consume_record(head, len32);
-      if (hdr_->spill_count) hdr_->spill_count -= 1;
+      consume_record(head, len32, kFlagSpill);
       throw;
     }
-    consume_record(head, len32);
-    if (hdr_->spill_count) hdr_->spill_count -= 1;
+    consume_record(head, len32, kFlagSpill);
   }
 
-  void consume_record(uint64_t head, uint32_t len32) {
+  void consume_record(uint64_t head, uint32_t len32, uint32_t flags = 0) {
     size_t need = record_bytes(len32);
-    hdr_->head = (head + need) % hdr_->capacity;
-    hdr_->used -= need;
-    hdr_->msg_count -= 1;
-    hdr_->total_out += 1;
+    uint64_t spill = hdr_->spill_count;
+    if ((flags & kFlagSpill) && spill) spill -= 1;
+    txn_arm(hdr_, (head + need) % hdr_->capacity, hdr_->tail,
+            hdr_->used - need, hdr_->msg_count - 1, hdr_->total_in,
+            hdr_->total_out + 1, spill);
+    txn_apply(hdr_);
   }
 
   void reclaim_dead_record(uint64_t head, uint32_t len32, uint32_t flags) {
     size_t need = record_bytes(len32);
-    hdr_->head = (head + need) % hdr_->capacity;
-    hdr_->used -= need;
-    hdr_->msg_count -= 1;
     // flags were written under the reserve lock, so they are valid even
     // though the payload is not: release the dead writer's spill budget
-    if ((flags & kFlagSpill) && hdr_->spill_count) hdr_->spill_count -= 1;
+    uint64_t spill = hdr_->spill_count;
+    if ((flags & kFlagSpill) && spill) spill -= 1;
+    txn_arm(hdr_, (head + need) % hdr_->capacity, hdr_->tail,
+            hdr_->used - need, hdr_->msg_count - 1, hdr_->total_in,
+            hdr_->total_out, spill);
+    txn_apply(hdr_);
   }
 
   // Shared blocking structure for recv / recv_into / peek.
@@ -662,7 +718,7 @@ class ShmRing {
       uint32_t snap = as_atomic(&hdr_->fut_not_empty)
                           ->load(std::memory_order_acquire);
       {
-        RobustLock lock(&hdr_->mu);
+        RobustLock lock(&hdr_->mu, hdr_);
         uint64_t head;
         uint32_t len32;
         uint32_t flags;
@@ -734,7 +790,7 @@ class ShmRing {
       uint32_t snap = as_atomic(&hdr_->fut_not_empty)
                           ->load(std::memory_order_acquire);
       {
-        RobustLock lock(&hdr_->mu);
+        RobustLock lock(&hdr_->mu, hdr_);
         uint64_t head;
         uint32_t len32;
         uint32_t flags;
@@ -826,13 +882,10 @@ class ShmRing {
             try {
               spill_read(nm, buf, real_len);
             } catch (...) {
-              consume_record(head, len32);
-              if (hdr_->spill_count) hdr_->spill_count -= 1;
+              consume_record(head, len32, kFlagSpill);
               throw;
             }
-            consume_record(head, len32);
-            // release the spill budget (same as consume_spill)
-            if (hdr_->spill_count) hdr_->spill_count -= 1;
+            consume_record(head, len32, kFlagSpill);
             result = (int64_t)real_len;
             return true;
           }
@@ -880,6 +933,7 @@ class ShmRing {
     deadline.tv_sec += 5;
     int rc = pthread_mutex_timedlock(&hdr_->mu, &deadline);
     if (rc == EOWNERDEAD) {
+      txn_recover(hdr_);
       pthread_mutex_consistent(&hdr_->mu);
       rc = 0;
     }
