@@ -21,7 +21,7 @@ degree the reference is ("one-line fiberization").
 
 import os
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from . import config as _config_mod
 from .config import Config  # noqa: F401
