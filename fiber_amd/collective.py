@@ -93,13 +93,14 @@ class GroupMaster:
 
     def descriptor(self, rank=None):
         self.ensure()
-        d = {
-            "gen": self.gen,
-            "world": self.world,
-            "host": self.host,
-            "port": self.port,
-            "backend": self.backend,
-        }
+        with self._lock:  # snapshot: a concurrent rebuild() must not
+            d = {         # yield a (new gen, old port) mix
+                "gen": self.gen,
+                "world": self.world,
+                "host": self.host,
+                "port": self.port,
+                "backend": self.backend,
+            }
         if rank is not None:
             d["rank"] = rank
         return d
