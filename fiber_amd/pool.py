@@ -421,6 +421,7 @@ def _execute_chunk(func, args, starmap, kwds):
 
 
 _current_worker_group = None
+_current_coll_state = None
 
 
 def current_worker_group():
@@ -449,8 +450,9 @@ class _WorkerCollState:
         self.result_sock = result_sock
         self.ident = ident
         self.maps = {}  # seq -> {shared, reduce, spec, partial}
-        global _current_worker_group
+        global _current_worker_group, _current_coll_state
         _current_worker_group = self.group
+        _current_coll_state = self  # introspection (tests/debug)
 
     # -- ctl processing ----------------------------------------------------
     def drain(self, timeout=0.0):
@@ -1137,6 +1139,10 @@ class ZPool:
                         )
                 elif seq in self._coll_seqs and self._inventory.done(seq):
                     self._coll_seqs.discard(seq)
+                    # free the workers' staged shared tensors (device
+                    # memory!) — without this every shared= map leaks
+                    # its broadcast copies in each worker forever
+                    self._ctl_broadcast(lambda rank, ident: ("drop", seq))
                 watcher = self._callback_watch.get(seq)
                 if watcher is not None:
                     done, error, results = self._inventory.peek(seq)

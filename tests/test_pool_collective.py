@@ -130,6 +130,26 @@ class TestRebuildOnDeath:
             pool.join()
 
 
+def _staged_maps_count():
+    import fiber_amd.pool as pm
+
+    return len(pm._current_coll_state.maps)
+
+
+class TestStagedCleanup:
+    def test_shared_tensors_freed_after_map(self, cpool):
+        """Completed collective maps must drop their staged shared
+        tensors in every worker (device memory on GPU pools) — the
+        master broadcasts a drop on completion."""
+        theta = torch.ones(64)
+        for _ in range(3):
+            cpool.map(_weighted, range(4), chunksize=1,
+                      shared={"theta": theta})
+        time.sleep(0.5)  # drop records are async to map completion
+        counts = cpool.run_on_all(_staged_maps_count)
+        assert counts == [0, 0], counts
+
+
 class TestReduceEdges:
     def test_empty_iterable_reduce(self, cpool):
         out = cpool.map(_onehot4, [], reduce="sum",
