@@ -45,8 +45,12 @@ def main():
     if args.pop_per_gpu is None:
         # Big shards are the MI355X-first choice (288 GB HBM, fixed costs
         # amortize, collectives stay large); per-GPU work is fixed as N
-        # grows (weak scaling).
-        args.pop_per_gpu = 16384 if args.model == "mlp" else 1024
+        # grows (weak scaling).  32768 members = 21.3 occupancy rounds of
+        # the rollout kernel (256 CU x 6 wg) — the 16384 shard wasted
+        # 6.7% in the last-round tail (measured ladder:
+        # profiles/r02_rollout_pmc.md: 83.6M @16k, 90.2M @32k, 91.2M
+        # @128k; 32k keeps ms/step driver-friendly).
+        args.pop_per_gpu = 32768 if args.model == "mlp" else 1024
     if args.horizon is None:
         args.horizon = 256 if args.model == "mlp" else 64
 
