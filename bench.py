@@ -50,7 +50,7 @@ def main():
         # 6.7% in the last-round tail (measured ladder:
         # profiles/r02_rollout_pmc.md: 83.6M @16k, 90.2M @32k, 91.2M
         # @128k; 32k keeps ms/step driver-friendly).
-        args.pop_per_gpu = 32768 if args.model == "mlp" else 1024
+        args.pop_per_gpu = 32768 if args.model == "mlp" else 2048
     if args.horizon is None:
         args.horizon = 256 if args.model == "mlp" else 64
 

@@ -104,7 +104,7 @@ def main():
         workers = args.workers or max(1, fiber_amd.gpu_count())
         cfg = dict(
             pop_per_gpu=args.pop_per_gpu
-            or (32768 if args.model == "mlp" else 1024),
+            or (32768 if args.model == "mlp" else 2048),
             horizon=args.horizon or (256 if args.model == "mlp" else 64),
         )
         stub = False
