@@ -52,6 +52,18 @@ class TestExamples:
 
 
 class TestCli:
+    def test_doctor(self):
+        proc = subprocess.run(
+            [sys.executable, "-m", "fiber_amd.cli", "doctor"],
+            capture_output=True,
+            text=True,
+            timeout=120,
+            cwd=ROOT,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr[-500:]
+        assert "shm ring self-test" in proc.stdout
+        assert "FAIL" not in proc.stdout
+
     def test_info(self):
         proc = subprocess.run(
             [sys.executable, "-m", "fiber_amd.cli", "info"],
