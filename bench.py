@@ -62,8 +62,21 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    device = torch.device("cuda", local_rank)
-    torch.cuda.set_device(device)
+    # FAM_BENCH_CPU=1: dress rehearsal of the EXACT driver invocation
+    # (torchrun rendezvous, env parsing, barriers, MAX-reduce, JSON
+    # contract) on a GPU-less box — HIP kernels stubbed, gloo backend.
+    # tests/test_bench_rehearsal.py runs it at world 2 so the 8-GPU
+    # SCALE path has been executed end-to-end before the driver tries.
+    cpu_rehearsal = bool(os.environ.get("FAM_BENCH_CPU"))
+    if cpu_rehearsal:
+        from fiber_amd import ops as _ops
+        from tests.test_es_distributed import _cpu_stub_ops
+
+        _cpu_stub_ops(_ops)
+        device = torch.device("cpu")
+    else:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
 
     ctx = None
     if world > 1:
@@ -73,7 +86,9 @@ def main():
         if os.environ.get("FAM_NCCL_DEBUG"):
             # RCCL-level diagnostics for ring/xGMI setup issues
             os.environ.setdefault("NCCL_DEBUG", "WARN")
-        ctx = RingContext(rank, world, backend="nccl", device=device)
+        ctx = RingContext(rank, world,
+                          backend="gloo" if cpu_rehearsal else "nccl",
+                          device=device)
         ctx.init()
 
     if args.model == "mlp":
@@ -92,18 +107,22 @@ def main():
         model_name = "es-conv-84x84x4-dqn-act6"
         dtype = "bf16+fp8(e4m3)"  # conv1/fc ride OCP fp8, rest bf16
 
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
     for i in range(args.warmup):
         engine.step(iteration=i)
 
     if ctx is not None:
         ctx.barrier()
-    torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for i in range(args.steps):
         engine.step(iteration=args.warmup + i)
     if ctx is not None:
         ctx.barrier()
-    torch.cuda.synchronize()
+    sync()
     elapsed = time.perf_counter() - t0
 
     if ctx is not None:
