@@ -20,3 +20,8 @@ tsan:  ## ThreadSanitizer run over the transport engine (no GPU needed)
 	g++ -std=c++17 -O1 -g -fsanitize=thread -DFAM_NO_PYBIND \
 	  fiber_amd/csrc/tsan_harness.cpp -o /tmp/fam_tsan -lpthread -lrt
 	/tmp/fam_tsan
+
+asan:  ## Address+UB sanitizer run over the transport engine
+	g++ -std=c++17 -O1 -g -fsanitize=address,undefined -DFAM_NO_PYBIND \
+	  fiber_amd/csrc/tsan_harness.cpp -o /tmp/fam_asan -lpthread -lrt
+	/tmp/fam_asan

@@ -161,3 +161,29 @@ class TestReduceEdges:
             cpool.map_async(_onehot4, range(4), reduce="sum",
                             reduce_spec=((4,), torch.float32),
                             callback=lambda r: None)
+
+
+def _slow_weighted(x, theta=None):
+    time.sleep(0.01)
+    return float(theta.sum()) * x
+
+
+class TestConcurrentCollectives:
+    def test_interleaved_shared_maps_and_spmd(self, cpool):
+        """Multiple in-flight collective maps + SPMD execs: the ctl
+        ordering guarantee must keep every rank's collective sequence
+        identical (a mismatch deadlocks or mixes stage broadcasts)."""
+        thetas = [torch.full((8,), float(k + 1)) for k in range(3)]
+        asyncs = [
+            cpool.map_async(_slow_weighted, range(6), chunksize=1,
+                            shared={"theta": thetas[k]})
+            for k in range(3)
+        ]
+        spmd = cpool.run_on_all(_group_allreduce_rank, timeout=120)
+        assert spmd == [3.0, 3.0]
+        for k, res in enumerate(asyncs):
+            want = [float(thetas[k].sum()) * x for x in range(6)]
+            assert res.get(120) == want
+        red = cpool.map(_onehot4, range(8), chunksize=2, reduce="sum",
+                        reduce_spec=((4,), torch.float32))
+        assert torch.equal(red, torch.full((4,), float(sum(range(8)))))
