@@ -41,6 +41,32 @@ def _pg_timeout():
     )
 
 
+_side_stream = None
+
+
+def stage_to_device(t, device):
+    """Host->device staging for shared-tensor fan-out: pinned buffer +
+    `hipMemcpyAsync` on a dedicated side stream (the north-star staging
+    path) — the copy engine runs the transfer while the
+    compute stream keeps working; the caller's stream only waits at the
+    end.  Device-resident and CPU-target tensors pass through."""
+    global _side_stream
+    import torch
+
+    if t.device == device:
+        return t.contiguous()
+    if device.type != "cuda" or t.device.type != "cpu":
+        return t.to(device).contiguous()
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream(device=device)
+    pinned = t.contiguous().pin_memory()
+    out = torch.empty_like(pinned, device=device)
+    with torch.cuda.stream(_side_stream):
+        out.copy_(pinned, non_blocking=True)
+    torch.cuda.current_stream().wait_stream(_side_stream)
+    return out
+
+
 class CollectiveError(RuntimeError):
     """A communicator-level failure (rendezvous / transport), as opposed
     to a user-code exception.  The pool master reacts by rotating the

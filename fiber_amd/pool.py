@@ -557,12 +557,16 @@ class _WorkerCollState:
         ONE host->device hand-off total instead of one per chunk."""
         import torch
 
+        from .collective import stage_to_device
+
         device = self.group.device
         shared = {}
         if self.group.rank == 0:
             src = serialization.loads(blob)
             for name in sorted(meta):
-                shared[name] = src[name].to(device).contiguous()
+                # pinned + hipMemcpyAsync on a side stream for host
+                # payloads; device tensors (IPC handles) pass through
+                shared[name] = stage_to_device(src[name], device)
         else:
             for name in sorted(meta):
                 shape, dtype_str = meta[name]

@@ -50,6 +50,22 @@ class TestPoolCollectiveGPU:
             pool.terminate()
             pool.join()
 
+    def test_shared_cpu_tensor_staged_pinned(self):
+        """A HOST tensor in shared= stages via the pinned side-stream
+        path (hipMemcpyAsync) and arrives device-resident."""
+        from fiber_amd.pool import ZPool
+
+        theta = torch.arange(4096, dtype=torch.float32)  # CPU
+        pool = ZPool(processes=1, gpu_per_worker=1, collective=True)
+        try:
+            out = pool.map(_dot_with_shared, range(4), chunksize=1,
+                           shared={"theta": theta})
+            want = [float(theta.sum()) * x for x in range(4)]
+            assert out == want
+        finally:
+            pool.terminate()
+            pool.join()
+
     def test_reduce_sum_on_device(self):
         from fiber_amd.pool import ZPool
 
