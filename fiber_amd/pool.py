@@ -1577,3 +1577,16 @@ class ResilientZPool(ZPool):
 
 
 Pool = ResilientZPool
+
+
+def __getattr__(name):  # PEP 562 lazy re-export
+    """``fiber_amd.pool.ClassicPool`` — the legacy two-queue pool
+    (reference fiber/pool.py:175-641 analog).  Lazy: classic_pool
+    imports THIS module for Inventory/_execute_chunk, so an eager
+    tail-import here is a circular import in any process that loads
+    classic_pool first (e.g. a worker unpickling _ClassicEntry)."""
+    if name == "ClassicPool":
+        from .classic_pool import ClassicPool
+
+        return ClassicPool
+    raise AttributeError("module %r has no attribute %r" % (__name__, name))
