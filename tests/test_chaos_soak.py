@@ -158,3 +158,48 @@ class TestRingChurn:
                     pass
             ring.close()
             ring.unlink()
+
+
+def _coll_allreduce_rank():
+    from fiber_amd.pool import current_worker_group
+
+    g = current_worker_group()
+    import torch
+
+    t = torch.tensor([float(g.rank + 1)])
+    g.allreduce(t)
+    return float(t[0])
+
+
+class TestCollectiveChurn:
+    def test_repeated_kill_recover_cycles(self, monkeypatch):
+        """Resilience x RCCL: murder a member of a collective pool,
+        verify the generation rotation recovers, repeat.  Every cycle
+        must converge to a working communicator (the failure policy's
+        whole claim)."""
+        monkeypatch.setenv("FAM_PG_TIMEOUT", "5")
+        from fiber_amd.pool import ZPool
+
+        cycles = 4 if os.environ.get("FAM_SOAK") else 2
+        pool = ZPool(processes=2, collective=True)
+        try:
+            assert pool.run_on_all(_coll_allreduce_rank,
+                                   timeout=120) == [3.0, 3.0]
+            for cycle in range(cycles):
+                with pool._worker_lock:
+                    victim = next(iter(pool._workers.values()))
+                victim.kill()
+                deadline = time.monotonic() + 90
+                while True:
+                    try:
+                        out = pool.run_on_all(_coll_allreduce_rank,
+                                              timeout=60)
+                        break
+                    except Exception:
+                        if time.monotonic() > deadline:
+                            raise
+                        time.sleep(0.5)
+                assert out == [3.0, 3.0], "cycle %d" % cycle
+        finally:
+            pool.terminate()
+            pool.join()
