@@ -87,12 +87,19 @@ class RingContext:
             timeout = datetime.timedelta(
                 seconds=float(os.environ.get("FAM_PG_TIMEOUT", "600"))
             )
+            kwargs = {}
+            if self.backend == "nccl" and self._device is not None \
+                    and self._device.type == "cuda":
+                # bind the communicator to this rank's MI355X up front
+                # (also silences the barrier device-guess warning)
+                kwargs["device_id"] = self._device
             try:
                 dist.init_process_group(
                     backend=self.backend,
                     rank=self.rank,
                     world_size=self.size,
                     timeout=timeout,
+                    **kwargs,
                 )
             except Exception:
                 import sys
