@@ -36,8 +36,14 @@ def _free_tcp_port():
 
 
 def _pg_timeout():
+    # Pool-group default is deliberately shorter than the Ring/bench
+    # default (600 s): recovery from a member death converges in a few
+    # timeout windows (a rank blocked on a dead generation must time out
+    # before it joins the new one), so this bounds the
+    # kill-to-recovered latency at ~2-3 minutes worst case.  RCCL init
+    # and per-block collectives finish in seconds when healthy.
     return datetime.timedelta(
-        seconds=float(os.environ.get("FAM_PG_TIMEOUT", "120"))
+        seconds=float(os.environ.get("FAM_PG_TIMEOUT", "60"))
     )
 
 
