@@ -139,7 +139,14 @@ class TestRingChurn:
             # AFTER an empty recv_many, re-exposing leftovers.)
             body = b"post-soak"
             sentinel = hashlib.blake2b(body, digest_size=8).digest() + body
-            ring.send(sentinel, 5.0)
+            # The ring may still be FULL of unconsumed leftovers, so the
+            # sentinel send needs the drain below to make space: send
+            # from a helper thread while this thread drains.
+            sent_ok = []
+            sender = threading.Thread(
+                target=lambda: sent_ok.append(ring.send(sentinel, 120.0))
+            )
+            sender.start()
             while True:
                 payload = ring.recv(10.0)
                 assert payload is not None, "ring wedged post-soak"
@@ -149,6 +156,8 @@ class TestRingChurn:
                 ).digest() == digest, "corrupt leftover after massacre"
                 if pbody == body:
                     break
+            sender.join(5)
+            assert sent_ok == [True]
         finally:
             for p in procs:
                 try:
