@@ -221,11 +221,9 @@ static void launch_conv_forward(uintptr_t wpert, uintptr_t w3_fp8,
                      (const unsigned char*)znoise, nenv,
                      (__hip_bfloat16*)act1);
   check(hipGetLastError(), "conv_layer1 launch");
-  // one wg per MEMBER (W2 staged once, envs looped inside) — grid is
-  // 16x smaller; the 8 concurrent stream chunks still fill the chip
-  hipLaunchKernelGGL(conv_layer2, dim3(nmembers), dim3(256), 0,
+  hipLaunchKernelGGL(conv_layer2, dim3(nenv), dim3(256), 0,
                      (hipStream_t)stream, (const __hip_bfloat16*)wpert,
-                     (const __hip_bfloat16*)act1, nmembers,
+                     (const __hip_bfloat16*)act1, nenv,
                      (unsigned char*)act2);
   check(hipGetLastError(), "conv_layer2 launch");
   hipLaunchKernelGGL(conv_fc, dim3(nmembers, 4), dim3(256), 0,

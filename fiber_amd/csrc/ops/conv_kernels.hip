@@ -311,82 +311,75 @@ conv_layer1(const __hip_bfloat16* __restrict__ wpert,
 }
 
 // ---------------------------------------------------------------------------
-// conv_layer2: act2 = tanh(conv(act1, W2) + b2); one wg per MEMBER, envs
-// looped.  M=32 (2 tiles), N=81 (6 tiles, padded), K=256 (8 tiles = 2
-// pixels each).  Staging W2 once per member-step instead of once per
-// (member, env) divides the dominant HBM term by E=16: the r02 PMC run
-// showed this kernel ~58% staging-bandwidth (W2 16 KB + act1 12.8 KB per
-// 22 us call) with nothing on-chip saturated.
+// conv_layer2: act2 = tanh(conv(act1, W2) + b2); one wg per (member, env).
+// M=32 (2 tiles), N=81 (6 tiles, padded), K=256 (8 tiles = 2 pixels each).
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(256)
 conv_layer2(const __hip_bfloat16* __restrict__ wpert,
-            const __hip_bfloat16* __restrict__ act1, int nmembers,
+            const __hip_bfloat16* __restrict__ act1, int nenv_total,
             unsigned char* __restrict__ act2) {
   __shared__ alignas(16) __hip_bfloat16 w2[C2][256];
   __shared__ float b2[C2];
   // act1 staged in LDS: the 4x4-stride-2 im2col pattern touches each
   // input pixel from up to 4 windows and BOTH M-tiles — ~7.7x read
-  // amplification that was going to L2/HBM (12.8 KB staged per env).
+  // amplification that was going to L2/HBM (12.8 KB staged once
+  // instead; w2 16 KB + act1 12.8 KB still leaves 4 wg/CU).
   __shared__ alignas(16) __hip_bfloat16 a1sh[O1 * O1 * C1];
-  const int member = blockIdx.x;
+  const int be = blockIdx.x;
+  const int member = be / CENV;
   const __hip_bfloat16* wm = wpert + (size_t)member * NP_CONV_PAD;
+  const __hip_bfloat16* in = act1 + (size_t)be * (O1 * O1 * C1);
+  unsigned char* out = act2 + (size_t)be * NFLAT;
 
   const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int kgrp = lane >> 4;
-
   for (int i = tid; i < C2 * 256 / 8; i += blockDim.x) {
     reinterpret_cast<bf16x8*>(&w2[0][0])[i] =
         reinterpret_cast<const bf16x8*>(wm + COFF_W2)[i];
   }
+  for (int i = tid; i < O1 * O1 * C1 / 8; i += blockDim.x)
+    reinterpret_cast<bf16x8*>(a1sh)[i] =
+        reinterpret_cast<const bf16x8*>(in)[i];
   if (tid < C2) b2[tid] = __bfloat162float(wm[COFF_B2 + tid]);
+  __syncthreads();
 
-  for (int e = 0; e < CENV; ++e) {
-    const size_t be = (size_t)member * CENV + e;
-    const __hip_bfloat16* in = act1 + be * (O1 * O1 * C1);
-    unsigned char* out = act2 + be * NFLAT;
-    __syncthreads();  // a1sh free (previous env's compute done)
-    for (int i = tid; i < O1 * O1 * C1 / 8; i += blockDim.x)
-      reinterpret_cast<bf16x8*>(a1sh)[i] =
-          reinterpret_cast<const bf16x8*>(in)[i];
-    __syncthreads();
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kgrp = lane >> 4;
 
-    // 12 tile jobs: job = mt*6 + ntile
-    for (int job = wave; job < 12; job += 4) {
-      const int mt = job / 6, nt = job % 6;
-      const int arow = mt * 16 + (lane & 15);
-      int pos = nt * 16 + (lane & 15);
-      const bool valid = pos < O2 * O2;
-      if (!valid) pos = 0;
-      const int oy = pos / O2, ox = pos % O2;
-      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  // 12 tile jobs: job = mt*6 + ntile
+  for (int job = wave; job < 12; job += 4) {
+    const int mt = job / 6, nt = job % 6;
+    const int arow = mt * 16 + (lane & 15);
+    int pos = nt * 16 + (lane & 15);
+    const bool valid = pos < O2 * O2;
+    if (!valid) pos = 0;
+    const int oy = pos / O2, ox = pos % O2;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        // K-tile kk covers pixels (ky = kk/2, kx = (kk%2)*2 + {0,1})
-        const int ky = kk >> 1;
-        const int kx0 = (kk & 1) * 2;
-        const int pix = ((oy * 2 + ky) * O1) + ox * 2 + kx0 + (kgrp >> 1);
-        const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            &w2[arow & 31][kk * 32 + kgrp * 8]);
-        const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &a1sh[pix * C1 + (kgrp & 1) * 8]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc,
-                                                      0, 0, 0);
-      }
-      const int drow = mt * 16 + kgrp * 4;
-      const int dpos = nt * 16 + (lane & 15);
-      if (dpos < O2 * O2) {
-        union {
-          unsigned char b[4];
-          uint32_t u;
-        } pk;
+    for (int kk = 0; kk < 8; ++kk) {
+      // K-tile kk covers pixels (ky = kk/2, kx = (kk%2)*2 + {0,1})
+      const int ky = kk >> 1;
+      const int kx0 = (kk & 1) * 2;
+      const int pix = ((oy * 2 + ky) * O1) + ox * 2 + kx0 + (kgrp >> 1);
+      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          &w2[arow & 31][kk * 32 + kgrp * 8]);
+      const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+          &a1sh[pix * C1 + (kgrp & 1) * 8]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0,
+                                                    0, 0);
+    }
+    const int drow = mt * 16 + kgrp * 4;
+    const int dpos = nt * 16 + (lane & 15);
+    if (dpos < O2 * O2) {
+      union {
+        unsigned char b[4];
+        uint32_t u;
+      } pk;
 #pragma unroll
-        for (int ri = 0; ri < 4; ++ri)
-          pk.b[ri] = __hip_fp8_e4m3(
-              fast_tanh_c(acc[ri] + b2[(drow + ri) & 31])).__x;
-        *reinterpret_cast<uint32_t*>(&out[dpos * C2 + (drow & 31)]) = pk.u;
-      }
+      for (int ri = 0; ri < 4; ++ri)
+        pk.b[ri] = __hip_fp8_e4m3(
+            fast_tanh_c(acc[ri] + b2[(drow + ri) & 31])).__x;
+      *reinterpret_cast<uint32_t*>(&out[dpos * C2 + (drow & 31)]) = pk.u;
     }
   }
 }
