@@ -28,6 +28,22 @@ def _gpu_onehot(x):
     return t
 
 
+def _slow_echo_gpu(x, theta=None):
+    import time
+
+    time.sleep(0.2)
+    return x + float(theta[0])
+
+
+def _gpu_group_probe():
+    from fiber_amd.pool import current_worker_group
+
+    g = current_worker_group()
+    t = torch.ones(4, device="cuda")
+    g.allreduce(t)  # world-1 RCCL collective on the NEW generation
+    return (g.rank, g.size, bool((t == 1).all().item()))
+
+
 def _engine_probe(k):
     from examples.es_pool import es_steps
 
@@ -78,6 +94,41 @@ class TestPoolCollectiveGPU:
             for x in range(16):
                 want[x % 8] += float(x)
             assert torch.equal(got.cpu(), want)
+        finally:
+            pool.terminate()
+            pool.join()
+
+    def test_kill_recover_world1_nccl(self, monkeypatch):
+        """The generation-rotation recovery on the REAL RCCL backend:
+        kill the GPU worker mid collective map, the map fails with
+        diagnosis, the respawned worker re-initializes a fresh NCCL
+        generation and SPMD work succeeds."""
+        import time
+
+        from fiber_amd.pool import ZPool
+
+        monkeypatch.setenv("FAM_PG_TIMEOUT", "30")
+        pool = ZPool(processes=1, gpu_per_worker=1, collective=True)
+        try:
+            theta = torch.ones(8, device="cuda")
+            res = pool.map_async(_slow_echo_gpu, range(40), chunksize=1,
+                                 shared={"theta": theta})
+            time.sleep(3.0)  # let the worker stage + start chunks
+            with pool._worker_lock:
+                victim = next(iter(pool._workers.values()))
+            victim.kill()
+            with pytest.raises(RuntimeError, match="collective"):
+                res.get(120)
+            deadline = time.monotonic() + 180
+            while True:
+                try:
+                    out = pool.run_on_all(_gpu_group_probe, timeout=90)
+                    break
+                except Exception:
+                    if time.monotonic() > deadline:
+                        raise
+                    time.sleep(1.0)
+            assert out == [(0, 1, True)]
         finally:
             pool.terminate()
             pool.join()
