@@ -1470,15 +1470,25 @@ class ZPool:
         # Workers killed by SIGTERM/SIGKILL never ran their atexit
         # cleanup; reap their per-ident reply rings (resilient mode)
         # master-side, including forked cores' derived idents.
+        import glob as _glob
+
         for ident in self._all_idents:
             names = [ident] + [
                 "%s:%d" % (ident, k) for k in range(1, self._nproc_per_job)
             ]
             for name in names:
+                base = "/dev/shm/%s.task.r.%s" % (self._name, name)
                 try:
-                    os.unlink("/dev/shm/%s.task.r.%s" % (self._name, name))
+                    os.unlink(base)
                 except OSError:
                     pass
+                # spill segments of the dead worker's reply ring (a
+                # SIGKILLed worker never ran its own sweep)
+                for path in _glob.glob(base + ".sp.*"):
+                    try:
+                        os.unlink(path)
+                    except OSError:
+                        pass
 
     def __enter__(self):
         return self
