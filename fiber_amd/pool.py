@@ -947,6 +947,8 @@ class ZPool:
 
     def _worker_loop(self):
         """Maintain the worker population; resubmit a dead worker's tasks."""
+        crash_streak = 0
+        crash_recv_mark = -1
         while self._state == "run":
             with self._worker_lock:
                 dead = [
@@ -963,6 +965,32 @@ class ZPool:
                 self._on_worker_death(ident, proc)
             if dead and self._collective:
                 self._rebuild_group([i for i, _ in dead])
+            if dead:
+                # Crash-loop detection: workers dying repeatedly while
+                # tasks are outstanding and NO results arrive means every
+                # respawn hits the same fatal error (e.g. the task
+                # function cannot be unpickled worker-side).  Without
+                # this, map() hangs forever; with it, outstanding maps
+                # fail with the dead worker's own log.
+                if self._recv == crash_recv_mark and self._sent > self._recv:
+                    crash_streak += len(dead)
+                else:
+                    crash_streak = len(dead)
+                    crash_recv_mark = self._recv
+                if crash_streak >= 2 * self._processes + 2:
+                    logs = ""
+                    try:
+                        logs = dead[-1][1].logs()[-1500:]
+                    except Exception:  # noqa: BLE001
+                        pass
+                    self._inventory.fail_all(RuntimeError(
+                        "pool workers are crash-looping (%d consecutive "
+                        "deaths with zero results while %d tasks are in "
+                        "flight); last worker log:\n%s"
+                        % (crash_streak, self._sent - self._recv, logs)
+                    ))
+                    crash_streak = 0
+                    crash_recv_mark = self._recv
             with self._worker_lock:
                 free = [s for s, sid in enumerate(self._slots)
                         if sid is None]

@@ -376,3 +376,33 @@ class TestAsyncCallbacks:
         finally:
             pool.terminate()
             pool.join()
+
+
+def _explode_on_load():
+    raise RuntimeError("boom on unpickle")
+
+
+class _ExplodingCallable:
+    """Pickles fine master-side; detonates when the worker unpickles it."""
+
+    def __reduce__(self):
+        return (_explode_on_load, ())
+
+    def __call__(self, x):  # pragma: no cover - never reached
+        return x
+
+
+class TestCrashLoopDetection:
+    def test_unpicklable_function_fails_instead_of_hanging(self):
+        """A function that cannot be unpickled worker-side kills every
+        respawned worker; the pool must diagnose the crash loop and fail
+        the map (with the worker's own log) instead of hanging forever
+        (found the expensive way: a heredoc-__main__ function hung a
+        GPU lease for its full timeout)."""
+        pool = ZPool(processes=2)
+        try:
+            with pytest.raises(RuntimeError, match="crash-looping"):
+                pool.map(_ExplodingCallable(), range(8), chunksize=1)
+        finally:
+            pool.terminate()
+            pool.join()
