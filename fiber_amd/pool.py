@@ -856,9 +856,15 @@ class ZPool:
                                  "worker process (cpu_per_worker == 1)")
             from .collective import GroupMaster
 
-            self._group_master = GroupMaster(
-                self._processes, collective_backend
+            # Backend follows the POOL's placement, not the machine:
+            # GPU-pinned workers (one device each) ride RCCL; CPU
+            # workers use gloo even on a GPU node (unpinned workers
+            # would otherwise all claim device 0 — NCCL forbids two
+            # ranks per device).
+            backend = collective_backend or (
+                "nccl" if gpu_per_worker else "gloo"
             )
+            self._group_master = GroupMaster(self._processes, backend)
 
         self._init_blob = None
         if initializer is not None:
