@@ -406,3 +406,30 @@ class TestCrashLoopDetection:
         finally:
             pool.terminate()
             pool.join()
+
+
+class TestDelayedBackend:
+    def test_pool_correct_with_slow_job_creation(self, monkeypatch):
+        """Reference DelayedBackend idiom (uber/fiber
+        tests/test_docker_backend.py:86-105): randomized create_job
+        delays must not affect pool correctness or ordering."""
+        import random as _random
+
+        from fiber_amd import backend as fam_backend
+        from fiber_amd.backends.local import Backend as LocalBackend
+
+        class DelayedBackend(LocalBackend):
+            def create_job(self, spec):
+                time.sleep(_random.uniform(0.0, 0.4))
+                return super().create_job(spec)
+
+        monkeypatch.setitem(fam_backend._backends, "local",
+                            DelayedBackend())
+        pool = ZPool(processes=4)
+        try:
+            assert pool.map(_square, range(100), chunksize=4) == [
+                x * x for x in range(100)
+            ]
+        finally:
+            pool.terminate()
+            pool.join()
